@@ -1,0 +1,199 @@
+"""Equivariant weight basis: Q_J intertwiners and per-edge basis kernels.
+
+Functional contract follows reference /root/reference/se3_transformer_pytorch/basis.py
+(get_basis at :153, basis_transformation_Q_J at :123) with these deliberate
+changes for the MI355X build:
+
+* Q_J is solved against our own least-squares Wigner-D (ops/wigner.py) — no
+  J_dense blobs, no lie_learn; sign fixed deterministically so every DP rank
+  computes bit-identical tables on CPU float64.
+* ``differentiable=True`` genuinely keeps the autograd graph through the
+  spherical harmonics (the reference's flag is inverted/defeated:
+  basis.py:171 vs :200-203).
+* Spherical harmonics for all J are evaluated in ONE vectorized pass straight
+  from cartesian offsets (no per-(l,m) recursion / global cache) — the same
+  structure as the fused HIP basis kernel.
+* A packed per-pair layout (`get_basis_packed`) is provided for the fused
+  conv kernels; `get_basis` returns the reference-shaped dict
+  (b, n, k, 1, 2*do+1, 1, 2*di+1, 2*min(di,do)+1).
+
+Disk persistence of Q_J is optional via the CACHE_PATH env var (reference
+basis.py:15-16 keeps the same env name); writes are atomic (tmp+rename), no
+file locks needed.
+"""
+from __future__ import annotations
+
+import os
+import tempfile
+from functools import lru_cache
+from itertools import product
+
+import torch
+
+from ..utils import to_order, torch_default_dtype
+from .sh import sh_packed_from_cartesian, sh_offset
+from .wigner import wigner_d
+
+__all__ = ['basis_transformation_Q_J', 'get_basis', 'get_basis_packed', 'get_R_tensor', 'num_basis_freq']
+
+# deterministic generic Euler angles for the intertwiner solve (any 5 generic
+# rotations characterize the intertwiner space; fixed seed => reproducible)
+_ANGLE_SEED = 271828
+
+
+def _solver_angles():
+    g = torch.Generator().manual_seed(_ANGLE_SEED)
+    return (torch.rand(5, 3, generator=g, dtype=torch.float64) * 6.28).tolist()
+
+
+def num_basis_freq(d_in: int, d_out: int) -> int:
+    return to_order(min(d_in, d_out))
+
+
+def get_R_tensor(order_out: int, order_in: int, a, b, c) -> torch.Tensor:
+    """Kronecker product D_out(R) ⊗ D_in(R) (reference basis.py:110)."""
+    return torch.kron(wigner_d(order_out, a, b, c), wigner_d(order_in, a, b, c))
+
+
+def _null_space_1d(mats, eps=1e-9) -> torch.Tensor:
+    """The single common null vector of the stacked matrices (f64 SVD)."""
+    a = torch.cat(mats, dim=0)
+    _, s, vh = torch.linalg.svd(a)
+    null = vh[s.shape[0] - 1:] if s[-1] < eps else vh[0:0]
+    # rows of vh beyond rank are the null space; with a 1-D null space the
+    # smallest singular value row is the solution
+    assert s[-1] < eps and (s.shape[0] < 2 or s[-2] > eps), \
+        f'intertwiner null space is not 1-dimensional (singular values tail: {s[-3:]})'
+    return vh[-1]
+
+
+def _cache_file():
+    path = os.environ.get('CACHE_PATH', os.path.expanduser('~/.cache/se3_transformer_amd'))
+    if os.environ.get('CLEAR_CACHE') is not None:
+        return None
+    return os.path.join(path, 'qj_tables_v1.pt')
+
+
+_qj_disk = None
+
+
+def _load_disk_cache():
+    global _qj_disk
+    if _qj_disk is not None:
+        return _qj_disk
+    _qj_disk = {}
+    f = _cache_file()
+    if f is not None and os.path.exists(f):
+        try:
+            _qj_disk = torch.load(f, weights_only=True)
+        except Exception:
+            _qj_disk = {}
+    return _qj_disk
+
+
+def _save_disk_cache():
+    f = _cache_file()
+    if f is None:
+        return
+    try:
+        os.makedirs(os.path.dirname(f), exist_ok=True)
+        fd, tmp = tempfile.mkstemp(dir=os.path.dirname(f))
+        os.close(fd)
+        torch.save(_qj_disk, tmp)
+        os.replace(tmp, f)
+    except OSError:
+        pass
+
+
+@lru_cache(maxsize=None)
+def basis_transformation_Q_J(J: int, order_in: int, order_out: int) -> torch.Tensor:
+    """Q_J: [ (2*order_out+1)*(2*order_in+1), 2J+1 ] float32 intertwiner with
+    (D_out ⊗ D_in) Q_J = Q_J D_J for every rotation (reference basis.py:123).
+    """
+    key = (J, order_in, order_out)
+    disk = _load_disk_cache()
+    if key in disk:
+        return disk[key]
+
+    with torch_default_dtype(torch.float64), torch.no_grad():
+        mats = []
+        for a, b, c in _solver_angles():
+            r_tensor = get_R_tensor(order_out, order_in, a, b, c)
+            d_j = wigner_d(J, a, b, c)
+            eye_r = torch.eye(r_tensor.shape[0], dtype=torch.float64)
+            eye_j = torch.eye(d_j.shape[0], dtype=torch.float64)
+            # vec(X): (R ⊗ I) vec - (I ⊗ D_J^T) vec = 0  <=>  R X = X D_J
+            mats.append(torch.kron(r_tensor.contiguous(), eye_j)
+                        - torch.kron(eye_r, d_j.t().contiguous()))
+        q = _null_space_1d(mats)
+        # deterministic sign: largest-|entry| component positive
+        idx = q.abs().argmax()
+        if q[idx] < 0:
+            q = -q
+        q = q.view(to_order(order_out) * to_order(order_in), to_order(J)).float().contiguous()
+
+    disk[key] = q
+    _save_disk_cache()
+    return q
+
+
+@lru_cache(maxsize=None)
+def _qj_transposed(J: int, d_in: int, d_out: int) -> torch.Tensor:
+    return basis_transformation_Q_J(J, d_in, d_out).t().contiguous()
+
+
+def _compute_sh(r_ij: torch.Tensor, max_J: int, differentiable: bool) -> torch.Tensor:
+    if differentiable:
+        return sh_packed_from_cartesian(max_J, r_ij)
+    with torch.no_grad():
+        return sh_packed_from_cartesian(max_J, r_ij)
+
+
+def get_basis(r_ij: torch.Tensor, max_degree: int, differentiable: bool = False):
+    """Reference-shaped equivariant basis dict.
+
+    Keys '{d_in},{d_out}' for d_in, d_out in 0..max_degree; values of shape
+    (*r_ij.shape[:-1], 1, 2*d_out+1, 1, 2*d_in+1, 2*min(d_in,d_out)+1)
+    (reference basis.py:153-205).
+    """
+    device, dtype = r_ij.device, r_ij.dtype
+    y_packed = _compute_sh(r_ij, 2 * max_degree, differentiable)
+
+    basis = {}
+    for d_in, d_out in product(range(max_degree + 1), range(max_degree + 1)):
+        k_js = []
+        for J in range(abs(d_in - d_out), d_in + d_out + 1):
+            q_t = _qj_transposed(J, d_in, d_out).to(device=device, dtype=dtype)
+            y_j = y_packed[..., sh_offset(J): sh_offset(J + 1)]
+            k_js.append(y_j @ q_t)  # [..., (2do+1)(2di+1)]
+        k = torch.stack(k_js, dim=-1)  # [..., (2do+1)(2di+1), F]
+        size = (*r_ij.shape[:-1], 1, to_order(d_out), 1, to_order(d_in),
+                num_basis_freq(d_in, d_out))
+        basis[f'{d_in},{d_out}'] = k.view(*size)
+
+    if not differentiable:
+        basis = {k: v.detach() for k, v in basis.items()}
+    return basis
+
+
+def get_basis_packed(r_ij: torch.Tensor, max_degree: int, differentiable: bool = False):
+    """Compact per-pair basis for the fused conv path.
+
+    Returns {(d_in, d_out): tensor [..., 2*d_out+1, 2*d_in+1, F]} with
+    F = 2*min(d_in,d_out)+1, contiguous, no broadcast singleton dims.
+    """
+    device, dtype = r_ij.device, r_ij.dtype
+    y_packed = _compute_sh(r_ij, 2 * max_degree, differentiable)
+
+    basis = {}
+    for d_in, d_out in product(range(max_degree + 1), range(max_degree + 1)):
+        k_js = []
+        for J in range(abs(d_in - d_out), d_in + d_out + 1):
+            q_t = _qj_transposed(J, d_in, d_out).to(device=device, dtype=dtype)
+            y_j = y_packed[..., sh_offset(J): sh_offset(J + 1)]
+            k_js.append(y_j @ q_t)
+        k = torch.stack(k_js, dim=-1)
+        k = k.view(*r_ij.shape[:-1], to_order(d_out), to_order(d_in),
+                   num_basis_freq(d_in, d_out))
+        basis[(d_in, d_out)] = k if differentiable else k.detach()
+    return basis

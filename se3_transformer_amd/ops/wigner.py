@@ -1,0 +1,153 @@
+"""Real Wigner-D matrices for SO(3), derived directly from spherical harmonics.
+
+The reference (/root/reference/se3_transformer_pytorch/irr_repr.py:12-30) loads
+precomputed ``J_dense`` block-diagonalization blobs and forms
+``D = X(a) J X(b) J X(c)``. Those blobs are not shipped here; instead D is
+obtained by *solving* the defining identity
+
+    Y_l(R x) = D_l(R) Y_l(x)
+
+as a float64 least-squares problem over a fixed, well-spread set of sample
+points on the sphere. This is exact up to f64 roundoff (~1e-13 relative,
+validated by the compose/irr_repr property test at 1e-10), self-consistent
+with our SH convention by construction, and needs no external data.
+
+Angle/axis conventions match the reference exactly:
+  * ``rot(a, b, c) = Rz(a) @ Ry(b) @ Rz(c)`` in standard coordinates,
+  * a point with spherical (alpha, beta) is ``Z(alpha) Y(beta) @ (0,0,1)``,
+  * ``spherical_harmonics(l, alpha, beta)`` evaluates at theta = pi - beta,
+    phi = alpha (irr_repr.py:103-104).
+"""
+from __future__ import annotations
+
+from functools import lru_cache
+from math import pi
+
+import torch
+
+from ..utils import cast_torch_tensor, to_order, torch_default_dtype
+from .sh import sh_packed_from_angles, sh_offset
+
+__all__ = [
+    'rot_z', 'rot_y', 'rot', 'x_to_alpha_beta', 'compose',
+    'spherical_harmonics', 'irr_repr', 'wigner_d',
+]
+
+
+@cast_torch_tensor
+def rot_z(gamma):
+    c, s = torch.cos(gamma), torch.sin(gamma)
+    one, zero = torch.ones_like(c), torch.zeros_like(c)
+    return torch.stack([
+        torch.stack([c, -s, zero]),
+        torch.stack([s, c, zero]),
+        torch.stack([zero, zero, one]),
+    ]).to(gamma.dtype)
+
+
+@cast_torch_tensor
+def rot_y(beta):
+    c, s = torch.cos(beta), torch.sin(beta)
+    one, zero = torch.ones_like(c), torch.zeros_like(c)
+    return torch.stack([
+        torch.stack([c, zero, s]),
+        torch.stack([zero, one, zero]),
+        torch.stack([-s, zero, c]),
+    ]).to(beta.dtype)
+
+
+def rot(alpha, beta, gamma):
+    """ZYZ Euler rotation, standard coordinates."""
+    return rot_z(alpha) @ rot_y(beta) @ rot_z(gamma)
+
+
+@cast_torch_tensor
+def x_to_alpha_beta(x):
+    """Spherical (alpha, beta) of a 3-vector: beta = acos(z/|x|), alpha = atan2(y, x)."""
+    x = x / torch.norm(x)
+    beta = torch.acos(x[2].clamp(-1., 1.))
+    alpha = torch.atan2(x[1], x[0])
+    return alpha, beta
+
+
+def compose(a1, b1, c1, a2, b2, c2):
+    """Euler angles of rot(a1,b1,c1) @ rot(a2,b2,c2)."""
+    comp = rot(a1, b1, c1) @ rot(a2, b2, c2)
+    xyz = comp @ torch.tensor([0., 0., 1.], dtype=comp.dtype)
+    a, b = x_to_alpha_beta(xyz)
+    rotz = rot(0, -b, -a) @ comp
+    c = torch.atan2(rotz[1, 0], rotz[0, 0])
+    return a, b, c
+
+
+def spherical_harmonics(order, alpha, beta):
+    """Y_order at the point Z(alpha) Y(beta) @ z-hat (reference irr_repr.py:103)."""
+    if not torch.is_tensor(alpha):
+        alpha = torch.tensor(alpha, dtype=torch.get_default_dtype())
+    if not torch.is_tensor(beta):
+        beta = torch.tensor(beta, dtype=torch.get_default_dtype())
+    return sh_packed_from_angles(order, pi - beta, alpha)[..., sh_offset(order):]
+
+
+@lru_cache(maxsize=None)
+def _sample_basis(order: int):
+    """Fixed sample points for degree `order` and the pseudo-inverse pieces.
+
+    Returns (pts [K,3] f64, Y [K, 2l+1] f64) where Y rows are Y_l at the points.
+    K = 4*(2l+1) quasi-random points keep the system well conditioned.
+    """
+    with torch_default_dtype(torch.float64):
+        n = to_order(order)
+        k = 4 * n
+        g = torch.Generator().manual_seed(918273 + order)
+        pts = torch.randn(k, 3, dtype=torch.float64, generator=g)
+        pts = pts / pts.norm(dim=-1, keepdim=True)
+        y = _sh_at_points(order, pts)
+    return pts, y
+
+
+def _sh_at_points(order: int, pts: torch.Tensor) -> torch.Tensor:
+    """Y_order at unit 3-vectors `pts` [K,3], standard-frame convention."""
+    beta = torch.acos(pts[:, 2].clamp(-1., 1.))
+    alpha = torch.atan2(pts[:, 1], pts[:, 0])
+    return sh_packed_from_angles(order, pi - beta, alpha)[..., sh_offset(order):]
+
+
+def _angles_key(v) -> float:
+    return float(v.item()) if torch.is_tensor(v) else float(v)
+
+
+@lru_cache(maxsize=4096)
+def _wigner_d_cached(order: int, a: float, b: float, c: float) -> torch.Tensor:
+    with torch_default_dtype(torch.float64):
+        pts, y = _sample_basis(order)
+        r = rot(torch.tensor(a, dtype=torch.float64),
+                torch.tensor(b, dtype=torch.float64),
+                torch.tensor(c, dtype=torch.float64))
+        # row-vector points: x' = R x  ->  pts' = pts @ R^T
+        y_rot = _sh_at_points(order, pts @ r.t())
+        # Y(Rx) = D Y(x):  y_rot = y @ D^T
+        sol = torch.linalg.lstsq(y, y_rot, driver='gelsd').solution
+    return sol.t().contiguous()
+
+
+def wigner_d_from_matrix(order: int, m3: torch.Tensor) -> torch.Tensor:
+    """Real Wigner-D for an arbitrary 3x3 rotation matrix (column convention:
+    x' = M x), float64, solved from Y(Mx) = D Y(x)."""
+    with torch_default_dtype(torch.float64):
+        pts, y = _sample_basis(order)
+        y_rot = _sh_at_points(order, pts @ m3.to(torch.float64).t())
+        sol = torch.linalg.lstsq(y, y_rot, driver='gelsd').solution
+    return sol.t().contiguous()
+
+
+def wigner_d(order: int, alpha, beta, gamma) -> torch.Tensor:
+    """Real Wigner-D matrix D_order(alpha, beta, gamma), float64, [2l+1, 2l+1]."""
+    return _wigner_d_cached(order, _angles_key(alpha), _angles_key(beta), _angles_key(gamma))
+
+
+def irr_repr(order: int, alpha, beta, gamma, dtype=None) -> torch.Tensor:
+    """Irreducible representation of SO(3) — compatible with `compose` and
+    `spherical_harmonics` (same contract as reference irr_repr.py:44-52)."""
+    dtype = dtype if dtype is not None else torch.get_default_dtype()
+    return wigner_d(order, alpha, beta, gamma).to(dtype)
