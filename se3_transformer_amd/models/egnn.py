@@ -1,0 +1,173 @@
+"""E(n)-GNN trunk generalized to higher-type features.
+
+Module tree parity with reference se3_transformer_pytorch.py:687-932
+(EGNN :707, EGnnNetwork :867). Functional rewrite: no in-place clamp_ or
+masked_fill_; the htype update einsum is a fused-kernel target.
+"""
+from __future__ import annotations
+
+import torch
+import torch.nn.functional as F
+from torch import nn
+
+from ..utils import batched_index_select, broadcat
+from .core import FeedForwardBlockSE3, HtypesNorm
+
+SiLU = nn.SiLU
+
+
+class EGNN(nn.Module):
+    def __init__(self, fiber, hidden_dim=32, edge_dim=0, init_eps=1e-3,
+                 coor_weights_clamp_value=None):
+        super().__init__()
+        self.fiber = fiber
+        node_dim = fiber[0]
+
+        htypes = [el for el in fiber if el.degrees != 0]
+        htype_dims = sum(el.dim for el in htypes)
+
+        edge_input_dim = node_dim * 2 + htype_dims + edge_dim + 1
+
+        self.node_norm = nn.LayerNorm(node_dim)
+
+        self.edge_mlp = nn.Sequential(
+            nn.Linear(edge_input_dim, edge_input_dim * 2),
+            SiLU(),
+            nn.Linear(edge_input_dim * 2, hidden_dim),
+            SiLU(),
+        )
+
+        self.htype_norms = nn.ModuleDict({})
+        self.htype_gating = nn.ModuleDict({})
+        for degree, dim in fiber:
+            if degree == 0:
+                continue
+            self.htype_norms[str(degree)] = HtypesNorm(dim)
+            self.htype_gating[str(degree)] = nn.Linear(node_dim, dim)
+
+        self.htypes_mlp = nn.Sequential(
+            nn.Linear(hidden_dim, hidden_dim * 4),
+            SiLU(),
+            nn.Linear(hidden_dim * 4, htype_dims),
+        )
+
+        self.node_mlp = nn.Sequential(
+            nn.Linear(node_dim + hidden_dim, node_dim * 2),
+            SiLU(),
+            nn.Linear(node_dim * 2, node_dim),
+        )
+
+        self.coor_weights_clamp_value = coor_weights_clamp_value
+        self.init_eps = init_eps
+        self.apply(self.init_)
+
+    def init_(self, module):
+        if type(module) in {nn.Linear}:
+            nn.init.normal_(module.weight, std=self.init_eps)
+
+    def forward(self, features, edge_info, rel_dist, mask=None, **kwargs):
+        neighbor_indices, neighbor_masks, edges = edge_info
+        mask = neighbor_masks
+
+        nodes = features['0'].squeeze(-1)  # b n d
+
+        htypes = [(deg, t) for deg, t in features.items() if deg != '0']
+        htype_degrees = [deg for deg, _ in htypes]
+        htype_dims = [t.shape[-2] for _, t in htypes]
+
+        # relative higher-type differences and their norms
+        rel_htypes = []
+        rel_htypes_dists = []
+        for _, htype in htypes:
+            rel_htype = htype.unsqueeze(2) - htype.unsqueeze(1)  # b i j d m
+            rel_htypes.append(rel_htype)
+            rel_htypes_dists.append(rel_htype.norm(dim=-1))
+
+        # edge MLP inputs
+        nodes_i = nodes.unsqueeze(2)  # b i 1 d
+        nodes_j = batched_index_select(nodes, neighbor_indices, dim=1)  # b i k d
+        neighbor_htype_dists = [batched_index_select(t, neighbor_indices, dim=2)
+                                for t in rel_htypes_dists]
+        coor_rel_dist = rel_dist.unsqueeze(-1)  # b i j 1
+
+        edge_mlp_inputs = broadcat((nodes_i, nodes_j, *neighbor_htype_dists, coor_rel_dist), dim=-1)
+        if edges is not None:
+            edge_mlp_inputs = torch.cat((edge_mlp_inputs, edges), dim=-1)
+
+        m_ij = self.edge_mlp(edge_mlp_inputs)
+
+        # higher-type updates
+        htype_weights = self.htypes_mlp(m_ij)
+        if self.coor_weights_clamp_value is not None:
+            c = self.coor_weights_clamp_value
+            htype_weights = htype_weights.clamp(min=-c, max=c)
+
+        if mask is not None:
+            htype_weights = htype_weights.masked_fill(~mask.unsqueeze(-1), 0.)
+
+        split_htype_weights = htype_weights.split(htype_dims, dim=-1)
+
+        htype_updates = []
+        for degree, rel_htype, htype_weight in zip(htype_degrees, rel_htypes, split_htype_weights):
+            normed_rel_htype = self.htype_norms[str(degree)](rel_htype)
+            normed_rel_htype = batched_index_select(normed_rel_htype, neighbor_indices, dim=2)
+            htype_updates.append(torch.einsum('bijdm,bijd->bidm', normed_rel_htype, htype_weight))
+
+        # node updates
+        if mask is not None:
+            m_ij = m_ij.masked_fill(~mask.unsqueeze(-1), 0.)
+        m_i = m_ij.sum(dim=-2)
+
+        normed_nodes = self.node_norm(nodes)
+        node_mlp_input = torch.cat((normed_nodes, m_i), dim=-1)
+        node_out = self.node_mlp(node_mlp_input) + nodes
+
+        out = dict(features)
+        out['0'] = node_out.unsqueeze(-1)
+
+        for degree, update in zip(htype_degrees, htype_updates):
+            out[degree] = out[degree] + update
+
+        for degree in htype_degrees:
+            gating = torch.sigmoid(self.htype_gating[str(degree)](node_out))
+            out[degree] = out[degree] * gating.unsqueeze(-1)
+
+        return out
+
+
+class EGnnNetwork(nn.Module):
+    def __init__(self, *, fiber, depth, edge_dim=0, hidden_dim=32,
+                 coor_weights_clamp_value=None, feedforward=False):
+        super().__init__()
+        self.fiber = fiber
+        self.layers = nn.ModuleList([])
+        for _ in range(depth):
+            self.layers.append(nn.ModuleList([
+                EGNN(fiber=fiber, edge_dim=edge_dim, hidden_dim=hidden_dim,
+                     coor_weights_clamp_value=coor_weights_clamp_value),
+                FeedForwardBlockSE3(fiber) if feedforward else None,
+            ]))
+
+    def forward(self, features, edge_info, rel_dist, basis, global_feats=None,
+                pos_emb=None, mask=None, **kwargs):
+        neighbor_indices, neighbor_masks, edges = edge_info
+        device = neighbor_indices.device
+
+        # prepend self-edges (EGNN attends to self; reference :901-913)
+        self_indices = torch.arange(neighbor_indices.shape[1], device=device)
+        self_indices = self_indices.view(1, -1, 1)
+        neighbor_indices = broadcat((self_indices, neighbor_indices), dim=-1)
+        neighbor_masks = F.pad(neighbor_masks, (1, 0), value=True)
+        rel_dist = F.pad(rel_dist, (1, 0), value=0.)
+        if edges is not None:
+            edges = F.pad(edges, (0, 0, 1, 0), value=0.)
+
+        edge_info = (neighbor_indices, neighbor_masks, edges)
+
+        for egnn, ff in self.layers:
+            features = egnn(features, edge_info=edge_info, rel_dist=rel_dist,
+                            basis=basis, global_feats=global_feats,
+                            pos_emb=pos_emb, mask=mask, **kwargs)
+            if ff is not None:
+                features = ff(features)
+        return features

@@ -1,0 +1,422 @@
+"""SE3Transformer — the user-facing model.
+
+Constructor signature (36 kwargs), forward contract and state-dict layout
+match the reference (se3_transformer_pytorch.py:936-1375). Internals are
+MI355X-first: packed basis computed in one pass, index-based (not
+masked_select) graph construction, fused conv path, no `splits` chunking.
+"""
+from __future__ import annotations
+
+import torch
+import torch.nn.functional as F
+from torch import nn
+
+from ..ops.basis import get_basis_packed
+from ..utils import (batched_index_select, cast_tuple, default, exists,
+                     masked_mean, map_values, to_order)
+from .attention import AttentionBlockSE3, AttentionSE3, OneHeadedKVAttentionSE3
+from .core import ConvSE3, FeedForwardBlockSE3, LinearSE3, NormSE3
+from .egnn import EGnnNetwork
+from .fiber import Fiber, FiberEl
+from .reversible import ReversibleSequence, SequentialSequence
+from .rotary import SinusoidalEmbeddings
+
+
+def _off_diag_indices(n: int, device):
+    """Column indices of the (n, n-1) 'all but self' layout: row i lists all
+    j != i in increasing order (replaces the reference's masked_select
+    reshape at :1224 with pure indexing)."""
+    cols = torch.arange(n - 1, device=device).unsqueeze(0).expand(n, n - 1)
+    rows = torch.arange(n, device=device).unsqueeze(1)
+    return cols + (cols >= rows).long()  # (n, n-1)
+
+
+def _remove_self(t, off_diag):
+    """Drop the diagonal of a (b, n, n, ...) tensor -> (b, n, n-1, ...)."""
+    b, n = t.shape[0], t.shape[1]
+    idx = off_diag.view(1, n, n - 1, *((1,) * (t.dim() - 3)))
+    idx = idx.expand(b, n, n - 1, *t.shape[3:])
+    return t.gather(2, idx)
+
+
+class SE3Transformer(nn.Module):
+    def __init__(
+        self,
+        *,
+        dim,
+        heads=8,
+        dim_head=24,
+        depth=2,
+        input_degrees=1,
+        num_degrees=None,
+        output_degrees=1,
+        valid_radius=1e5,
+        reduce_dim_out=False,
+        num_tokens=None,
+        num_positions=None,
+        num_edge_tokens=None,
+        edge_dim=None,
+        reversible=False,
+        attend_self=True,
+        use_null_kv=False,
+        differentiable_coors=False,
+        fourier_encode_dist=False,
+        rel_dist_num_fourier_features=4,
+        num_neighbors=float('inf'),
+        attend_sparse_neighbors=False,
+        num_adj_degrees=None,
+        adj_dim=0,
+        max_sparse_neighbors=float('inf'),
+        dim_in=None,
+        dim_out=None,
+        norm_out=False,
+        num_conv_layers=0,
+        causal=False,
+        splits=4,
+        global_feats_dim=None,
+        linear_proj_keys=False,
+        one_headed_key_values=False,
+        tie_key_values=False,
+        rotary_position=False,
+        rotary_rel_dist=False,
+        norm_gated_scale=False,
+        use_egnn=False,
+        egnn_hidden_dim=32,
+        egnn_weights_clamp_value=None,
+        egnn_feedforward=False,
+        hidden_fiber_dict=None,
+        out_fiber_dict=None,
+    ):
+        super().__init__()
+        dim_in = default(dim_in, dim)
+        self.dim_in = cast_tuple(dim_in, input_degrees)
+        self.dim = dim
+
+        # embeddings
+        self.token_emb = nn.Embedding(num_tokens, dim) if exists(num_tokens) else None
+        self.num_positions = num_positions
+        self.pos_emb = nn.Embedding(num_positions, dim) if exists(num_positions) else None
+
+        self.rotary_rel_dist = rotary_rel_dist
+        self.rotary_position = rotary_position
+        self.rotary_pos_emb = None
+        if rotary_position or rotary_rel_dist:
+            num_rotaries = int(rotary_position) + int(rotary_rel_dist)
+            self.rotary_pos_emb = SinusoidalEmbeddings(dim_head // num_rotaries)
+
+        # edges
+        assert not (exists(num_edge_tokens) and not exists(edge_dim)), \
+            'edge_dim must be supplied if SE3 transformer is to have edge tokens'
+        self.edge_emb = nn.Embedding(num_edge_tokens, edge_dim) if exists(num_edge_tokens) else None
+        self.has_edges = exists(edge_dim) and edge_dim > 0
+
+        self.input_degrees = input_degrees
+
+        assert not (exists(num_adj_degrees) and num_adj_degrees < 1), \
+            'num_adj_degrees must be at least 1'
+
+        self.num_degrees = num_degrees if exists(num_degrees) else (max(hidden_fiber_dict.keys()) + 1)
+
+        output_degrees = output_degrees if not use_egnn else None
+        self.output_degrees = output_degrees
+
+        self.differentiable_coors = differentiable_coors
+
+        # neighbor hyperparameters
+        self.valid_radius = valid_radius
+        self.num_neighbors = num_neighbors
+
+        self.attend_sparse_neighbors = attend_sparse_neighbors
+        self.max_sparse_neighbors = max_sparse_neighbors
+
+        self.num_adj_degrees = num_adj_degrees
+        self.adj_emb = nn.Embedding(num_adj_degrees + 1, adj_dim) \
+            if exists(num_adj_degrees) and adj_dim > 0 else None
+
+        edge_dim = (edge_dim if self.has_edges else 0) + (adj_dim if exists(self.adj_emb) else 0)
+
+        # fibers
+        dim_in = default(dim_in, dim)
+        dim_out = default(dim_out, dim)
+        assert exists(num_degrees) or exists(hidden_fiber_dict), \
+            'either num_degrees or hidden_fiber_dict must be specified'
+
+        fiber_in = Fiber.create(input_degrees, dim_in)
+        if exists(hidden_fiber_dict):
+            fiber_hidden = Fiber(hidden_fiber_dict)
+        elif exists(num_degrees):
+            fiber_hidden = Fiber.create(num_degrees, dim)
+
+        if exists(out_fiber_dict):
+            fiber_out = Fiber(out_fiber_dict)
+            self.output_degrees = max(out_fiber_dict.keys()) + 1
+        elif exists(output_degrees):
+            fiber_out = Fiber.create(output_degrees, dim_out)
+        else:
+            fiber_out = None
+
+        conv_kwargs = dict(edge_dim=edge_dim, fourier_encode_dist=fourier_encode_dist,
+                           num_fourier_features=rel_dist_num_fourier_features, splits=splits)
+
+        assert not (causal and not attend_self), 'attend_self must be on in causal mode'
+        self.causal = causal
+
+        self.conv_in = ConvSE3(fiber_in, fiber_hidden, **conv_kwargs)
+
+        self.convs = nn.ModuleList([])
+        for _ in range(num_conv_layers):
+            self.convs.append(nn.ModuleList([
+                ConvSE3(fiber_hidden, fiber_hidden, **conv_kwargs),
+                NormSE3(fiber_hidden, gated_scale=norm_gated_scale),
+            ]))
+
+        self.accept_global_feats = exists(global_feats_dim)
+        assert not (reversible and self.accept_global_feats), \
+            'reversibility and global features are not compatible'
+
+        # trunk
+        self.attend_self = attend_self
+        default_attention_klass = OneHeadedKVAttentionSE3 if one_headed_key_values else AttentionSE3
+
+        if use_egnn:
+            self.net = EGnnNetwork(fiber=fiber_hidden, depth=depth, edge_dim=edge_dim,
+                                   hidden_dim=egnn_hidden_dim,
+                                   coor_weights_clamp_value=egnn_weights_clamp_value,
+                                   feedforward=egnn_feedforward)
+        else:
+            layers = nn.ModuleList([])
+            for _ in range(depth):
+                layers.append(nn.ModuleList([
+                    AttentionBlockSE3(fiber_hidden, heads=heads, dim_head=dim_head,
+                                      attend_self=attend_self, edge_dim=edge_dim,
+                                      fourier_encode_dist=fourier_encode_dist,
+                                      rel_dist_num_fourier_features=rel_dist_num_fourier_features,
+                                      use_null_kv=use_null_kv, splits=splits,
+                                      global_feats_dim=global_feats_dim,
+                                      linear_proj_keys=linear_proj_keys,
+                                      attention_klass=default_attention_klass,
+                                      tie_key_values=tie_key_values,
+                                      norm_gated_scale=norm_gated_scale),
+                    FeedForwardBlockSE3(fiber_hidden, norm_gated_scale=norm_gated_scale),
+                ]))
+            execution_class = ReversibleSequence if reversible else SequentialSequence
+            self.net = execution_class(layers)
+
+        # out
+        self.conv_out = ConvSE3(fiber_hidden, fiber_out, **conv_kwargs) if exists(fiber_out) else None
+        self.norm = NormSE3(fiber_out, gated_scale=norm_gated_scale, nonlin=nn.Identity()) \
+            if (norm_out or reversible) and exists(fiber_out) else nn.Identity()
+
+        final_fiber = default(fiber_out, fiber_hidden)
+        self.linear_out = LinearSE3(
+            final_fiber,
+            Fiber([FiberEl(degrees=deg, dim=1) for deg, _ in final_fiber]),
+        ) if reduce_dim_out else None
+
+    def forward(self, feats, coors, mask=None, adj_mat=None, edges=None,
+                return_type=None, return_pooled=False, neighbor_mask=None,
+                global_feats=None):
+        assert not (self.accept_global_feats ^ exists(global_feats)), \
+            'global features must be passed iff the model was built with global_feats_dim'
+
+        _mask = mask
+
+        if self.output_degrees == 1:
+            return_type = 0
+
+        if exists(self.token_emb):
+            feats = self.token_emb(feats)
+
+        if exists(self.pos_emb):
+            assert feats.shape[1] <= self.num_positions, \
+                'sequence length must not exceed num_positions'
+            pos_emb = self.pos_emb(torch.arange(feats.shape[1], device=feats.device))
+            feats = feats + pos_emb.unsqueeze(0)
+
+        assert not (self.attend_sparse_neighbors and not exists(adj_mat)), \
+            'adjacency matrix must be passed in when attend_sparse_neighbors is on'
+        assert not (self.has_edges and not exists(edges)), \
+            'edge tokens/values must be supplied when edge_dim was given'
+
+        if torch.is_tensor(feats):
+            feats = {'0': feats[..., None]}
+        if torch.is_tensor(global_feats):
+            global_feats = {'0': global_feats[..., None]}
+
+        b, n, d = feats['0'].shape[:3]
+        device = feats['0'].device
+
+        assert d == self.dim_in[0], \
+            f'feature dimension {d} must match dimension given at init {self.dim_in[0]}'
+        assert set(map(int, feats.keys())) == set(range(self.input_degrees)), \
+            f'input must have degrees 0..{self.input_degrees - 1}'
+
+        num_degrees = self.num_degrees
+        neighbors = self.num_neighbors
+        max_sparse_neighbors = self.max_sparse_neighbors
+        valid_radius = self.valid_radius
+
+        assert self.attend_sparse_neighbors or neighbors > 0, \
+            'must attend to sparse neighbors or have num_neighbors > 0'
+
+        off_diag = _off_diag_indices(n, device)  # (n, n-1)
+
+        # N-hop adjacency labels
+        adj_indices = None
+        if exists(self.num_adj_degrees):
+            if adj_mat.dim() == 2:
+                adj_mat = adj_mat.unsqueeze(0).expand(b, n, n).clone()
+            adj_indices = adj_mat.long()
+            for ind in range(self.num_adj_degrees - 1):
+                degree = ind + 2
+                next_adj = (adj_mat.float() @ adj_mat.float()) > 0
+                new_hop = next_adj & ~adj_mat.bool()
+                adj_indices = adj_indices.masked_fill(new_hop, degree)
+                adj_mat = next_adj
+            adj_indices = _remove_self(adj_indices, off_diag)
+
+        # sparse neighbors from adjacency
+        sparse_neighbor_mask = None
+        num_sparse_neighbors = 0
+        if self.attend_sparse_neighbors:
+            assert exists(adj_mat), 'adjacency matrix must be passed in'
+            if adj_mat.dim() == 2:
+                adj_mat = adj_mat.unsqueeze(0).expand(b, n, n)
+            adj_mat = _remove_self(adj_mat, off_diag)
+            adj_values = adj_mat.float()
+            adj_max_neighbors = int(adj_values.sum(dim=-1).max().item())
+            if max_sparse_neighbors < adj_max_neighbors:
+                adj_values = adj_values + torch.empty_like(adj_values).uniform_(-0.01, 0.01)
+            num_sparse_neighbors = int(min(max_sparse_neighbors, adj_max_neighbors))
+            values, indices = adj_values.topk(num_sparse_neighbors, dim=-1)
+            sparse_neighbor_mask = torch.zeros_like(adj_values).scatter_(-1, indices, values) > 0.5
+
+        # relative geometry (self excluded by indexing, not masked_select)
+        indices = off_diag.unsqueeze(0).expand(b, n, n - 1)
+        rel_pos_full = coors.unsqueeze(2) - coors.unsqueeze(1)  # b i j 3
+        rel_pos = _remove_self(rel_pos_full, off_diag)          # b i j-1 3
+
+        if exists(mask):
+            mask_pair = mask.unsqueeze(2) & mask.unsqueeze(1)
+            mask_pair = _remove_self(mask_pair, off_diag)
+
+        if exists(edges):
+            if exists(self.edge_emb):
+                edges = self.edge_emb(edges)
+            edges = _remove_self(edges, off_diag)
+
+        if exists(self.adj_emb):
+            adj_emb = self.adj_emb(adj_indices)
+            edges = torch.cat((edges, adj_emb), dim=-1) if exists(edges) else adj_emb
+
+        rel_dist = rel_pos.norm(dim=-1)
+
+        # neighbor selection distances
+        modified_rel_dist = rel_dist
+        max_value = torch.finfo(modified_rel_dist.dtype).max
+
+        if exists(neighbor_mask):
+            neighbor_mask = _remove_self(neighbor_mask, off_diag)
+            max_neighbors = int(neighbor_mask.sum(dim=-1).max().item())
+            if max_neighbors > neighbors:
+                print(f'neighbor_mask shows maximum number of neighbors as {max_neighbors} '
+                      f'but specified number of neighbors is {neighbors}')
+            modified_rel_dist = modified_rel_dist.masked_fill(~neighbor_mask, max_value)
+
+        if exists(sparse_neighbor_mask):
+            modified_rel_dist = modified_rel_dist.masked_fill(sparse_neighbor_mask, 0.)
+
+        if self.causal:
+            causal_mask = torch.ones(n, n - 1, device=device).triu().bool()
+            modified_rel_dist = modified_rel_dist.masked_fill(causal_mask.unsqueeze(0), max_value)
+
+        if neighbors == 0:
+            valid_radius = 0
+
+        neighbors = int(min(neighbors, n - 1))
+        total_neighbors = int(neighbors + num_sparse_neighbors)
+        assert total_neighbors > 0, 'must be fetching at least 1 neighbor'
+        total_neighbors = int(min(total_neighbors, n - 1))
+
+        dist_values, nearest_indices = modified_rel_dist.topk(total_neighbors, dim=-1, largest=False)
+        neighbor_mask = dist_values <= valid_radius
+
+        neighbor_rel_dist = batched_index_select(rel_dist, nearest_indices, dim=2)
+        neighbor_rel_pos = batched_index_select(rel_pos, nearest_indices, dim=2)
+        neighbor_indices = batched_index_select(indices, nearest_indices, dim=2)
+
+        if exists(mask):
+            neighbor_mask = neighbor_mask & batched_index_select(mask_pair, nearest_indices, dim=2)
+
+        if exists(edges):
+            edges = batched_index_select(edges, nearest_indices, dim=2)
+
+        # rotary embeddings
+        rotary_pos_emb = None
+        rotary_query_pos_emb = None
+        rotary_key_pos_emb = None
+
+        if self.rotary_position:
+            seq = torch.arange(n, device=device)
+            seq_pos_emb = self.rotary_pos_emb(seq)
+            self_indices = torch.arange(neighbor_indices.shape[1], device=device)
+            self_indices = self_indices.view(1, -1, 1).expand(b, -1, 1)
+            neighbor_indices_with_self = torch.cat((self_indices, neighbor_indices), dim=2)
+            pos_emb = batched_index_select(seq_pos_emb, neighbor_indices_with_self, dim=0)
+            rotary_key_pos_emb = pos_emb
+            rotary_query_pos_emb = seq_pos_emb.unsqueeze(0).expand(b, *seq_pos_emb.shape)
+
+        if self.rotary_rel_dist:
+            neighbor_rel_dist_with_self = F.pad(neighbor_rel_dist, (1, 0), value=0) * 1e2
+            rel_dist_pos_emb = self.rotary_pos_emb(neighbor_rel_dist_with_self)
+            rotary_key_pos_emb = rel_dist_pos_emb if rotary_key_pos_emb is None \
+                else torch.cat((rotary_key_pos_emb, rel_dist_pos_emb), dim=-1)
+
+            query_dist = torch.zeros(n, device=device)
+            query_pos_emb = self.rotary_pos_emb(query_dist)
+            query_pos_emb = query_pos_emb.unsqueeze(0).expand(b, *query_pos_emb.shape)
+            rotary_query_pos_emb = query_pos_emb if rotary_query_pos_emb is None \
+                else torch.cat((rotary_query_pos_emb, query_pos_emb), dim=-1)
+
+        if exists(rotary_query_pos_emb) and exists(rotary_key_pos_emb):
+            rotary_pos_emb = (rotary_query_pos_emb, rotary_key_pos_emb)
+
+        # equivariant basis (packed layout; one fused SH pass)
+        basis = get_basis_packed(neighbor_rel_pos, num_degrees - 1,
+                                 differentiable=self.differentiable_coors)
+
+        # main trunk
+        edge_info = (neighbor_indices, neighbor_mask, edges)
+        x = feats
+
+        x = self.conv_in(x, edge_info, rel_dist=neighbor_rel_dist, basis=basis)
+
+        for conv, nonlin in self.convs:
+            x = nonlin(x)
+            x = conv(x, edge_info, rel_dist=neighbor_rel_dist, basis=basis)
+
+        x = self.net(x, edge_info=edge_info, rel_dist=neighbor_rel_dist,
+                     basis=basis, global_feats=global_feats,
+                     pos_emb=rotary_pos_emb, mask=_mask)
+
+        if exists(self.conv_out):
+            x = self.conv_out(x, edge_info, rel_dist=neighbor_rel_dist, basis=basis)
+
+        x = self.norm(x)
+
+        if exists(self.linear_out):
+            x = self.linear_out(x)
+            x = map_values(lambda t: t.squeeze(dim=2), x)
+
+        if return_pooled:
+            mask_fn = (lambda t: masked_mean(t, _mask, dim=1)) if exists(_mask) \
+                else (lambda t: t.mean(dim=1))
+            x = map_values(mask_fn, x)
+
+        if '0' in x:
+            x['0'] = x['0'].squeeze(dim=-1)
+
+        if exists(return_type):
+            return x[str(return_type)]
+
+        return x
