@@ -1,0 +1,160 @@
+#!/usr/bin/env python
+"""Flagship training benchmark: SE(3)-Transformer denoising-style step.
+
+Metric (BASELINE.json): train samples/sec (fwd+bwd+optimizer), 1024-point
+clouds, dim=512, heads=8, dim_head=64, depth=6, num_degrees=4, bf16 compute,
+at 1/2/4/8 GPUs (weak scaling: per-GPU batch fixed). Synthetic point clouds,
+random-init weights (no network access for datasets).
+
+Launch (driver contract):
+  python bench.py --gpus 1 --steps K --warmup W            # single GPU/CPU
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+from se3_transformer_amd import SE3Transformer
+from se3_transformer_amd.parallel import DistributedDataParallelSE3, setup_distributed
+
+
+def build_model(args, device):
+    with torch.device(device):
+        model = SE3Transformer(
+            dim=args.dim,
+            heads=args.heads,
+            dim_head=args.dim_head,
+            depth=args.depth,
+            num_degrees=args.num_degrees,
+            valid_radius=args.valid_radius,
+            num_neighbors=args.num_neighbors,
+            attend_self=True,
+        )
+    return model
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument('--gpus', type=int, default=1)
+    p.add_argument('--steps', type=int, default=5)
+    p.add_argument('--warmup', type=int, default=2)
+    p.add_argument('--batch', type=int, default=1, help='per-GPU batch size')
+    p.add_argument('--points', type=int, default=1024)
+    p.add_argument('--dim', type=int, default=512)
+    p.add_argument('--heads', type=int, default=8)
+    p.add_argument('--dim-head', type=int, default=64)
+    p.add_argument('--depth', type=int, default=6)
+    p.add_argument('--num-degrees', type=int, default=4)
+    p.add_argument('--num-neighbors', type=int, default=8)
+    p.add_argument('--valid-radius', type=float, default=10.)
+    p.add_argument('--dtype', type=str, default='bf16', choices=['bf16', 'fp32'])
+    p.add_argument('--device', type=str, default=None)
+    args = p.parse_args()
+
+    rank, world, local_rank = setup_distributed()
+    use_cuda = torch.cuda.is_available()
+    if args.device:
+        device = torch.device(args.device)
+    else:
+        device = torch.device(f'cuda:{local_rank}') if use_cuda else torch.device('cpu')
+    if use_cuda:
+        torch.cuda.set_device(device)
+
+    torch.manual_seed(1234)  # identical init on every rank (bypasses broadcast cost)
+
+    model = build_model(args, device)
+    n_params = sum(p_.numel() for p_ in model.parameters())
+
+    ddp = DistributedDataParallelSE3(model) if world > 1 else None
+    runner = ddp if ddp is not None else model
+
+    opt = torch.optim.SGD(model.parameters(), lr=1e-4)
+
+    # synthetic protein-like point clouds, fixed per rank
+    g = torch.Generator(device='cpu').manual_seed(1000 + rank)
+    feats = torch.randn(args.batch, args.points, args.dim, generator=g).to(device)
+    coors = (torch.randn(args.batch, args.points, 3, generator=g) * 2.0).to(device)
+    mask = torch.ones(args.batch, args.points, dtype=torch.bool, device=device)
+    target = torch.randn(args.batch, args.points, args.dim, generator=g).to(device)
+
+    autocast_dtype = torch.bfloat16 if args.dtype == 'bf16' else torch.float32
+    autocast_enabled = args.dtype == 'bf16'
+
+    def train_step():
+        if ddp is not None:
+            ddp.zero_grad_buffers()
+        else:
+            opt.zero_grad(set_to_none=True)
+        with torch.autocast(device_type=device.type, dtype=autocast_dtype,
+                            enabled=autocast_enabled):
+            out = runner(feats, coors, mask, return_type=0)
+            loss = (out.float() - target).pow(2).mean()
+        loss.backward()
+        if ddp is not None:
+            ddp.finalize()
+        opt.step()
+        return loss
+
+    def barrier_sync():
+        if world > 1:
+            torch.distributed.barrier()
+        if use_cuda:
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        train_step()
+
+    barrier_sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        train_step()
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks (slowest rank defines throughput)
+    if world > 1:
+        t = torch.tensor([elapsed], device=device if use_cuda else 'cpu')
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = t.item()
+
+    ms_per_step = elapsed / args.steps * 1000.
+    samples_per_sec = (args.batch * world * args.steps) / elapsed
+
+    if rank == 0:
+        print(json.dumps({
+            'metric': 'train samples/sec (fwd+bwd), 1024-pt dim=512 depth=6 num_degrees=4',
+            'value': samples_per_sec,
+            'unit': 'samples/sec',
+            'n_gpus': world,
+            'steps': args.steps,
+            'warmup': args.warmup,
+            'ms_per_step': ms_per_step,
+            'higher_is_better': True,
+            'scaling': 'weak',
+            'vs_baseline': None,
+            'dtype': args.dtype,
+            'data': 'synthetic',
+            'config': {
+                'model': 'SE3Transformer',
+                'dim': args.dim, 'heads': args.heads, 'dim_head': args.dim_head,
+                'depth': args.depth, 'num_degrees': args.num_degrees,
+                'points': args.points, 'num_neighbors': args.num_neighbors,
+                'global_batch': args.batch * world,
+                'parallelism': f'dp{world}',
+                'n_params': n_params,
+                'optimizer': 'sgd',
+            },
+        }))
+
+    if world > 1:
+        torch.distributed.destroy_process_group()
+
+
+if __name__ == '__main__':
+    main()

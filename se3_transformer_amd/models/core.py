@@ -205,14 +205,27 @@ class PairwiseConv(nn.Module):
         per_edge = mo * mi * F_ * elem_size
         chunk = max(1, min(e_total, max_chunk_bytes // max(per_edge, 1)))
 
+        def run_chunk(ef_c, bx_c, xg_c):
+            r = self.rp(ef_c).to(xg_c.dtype)                     # (E, mo, mi, F)
+            u = torch.einsum('eoif,eci->eocf', bx_c, xg_c)       # (E, O, mi, F)
+            return torch.matmul(
+                r.reshape(-1, mo, mi * F_),
+                u.permute(0, 2, 3, 1).reshape(-1, mi * F_, O))   # (E, mo, O)
+
+        # The per-chunk radial activation (E_chunk * mo * mi * F) is the
+        # memory hog; when training a large config, recompute it in backward
+        # (activation checkpointing) instead of keeping every chunk alive —
+        # the same recompute-R strategy the fused HIP backward uses.
+        use_ckpt = torch.is_grad_enabled() and (e_total * per_edge > (1 << 26))
+
         outs = []
         for s in range(0, e_total, chunk):
             e = slice(s, s + chunk)
-            r = self.rp(ef[e]).to(xg.dtype)                      # (E, mo, mi, F)
-            u = torch.einsum('eoif,eci->eocf', bx[e], xg[e])     # (E, O, mi, F)
-            out = torch.matmul(
-                r.reshape(-1, mo, mi * F_),
-                u.permute(0, 2, 3, 1).reshape(-1, mi * F_, O))   # (E, mo, O)
+            if use_ckpt:
+                out = torch.utils.checkpoint.checkpoint(
+                    run_chunk, ef[e], bx[e], xg[e], use_reentrant=False)
+            else:
+                out = run_chunk(ef[e], bx[e], xg[e])
             outs.append(out)
         out = torch.cat(outs, dim=0) if len(outs) > 1 else outs[0]
         return out.view(*lead, mo, O)

@@ -1,0 +1,1 @@
+from .ddp import DistributedDataParallelSE3, setup_distributed
