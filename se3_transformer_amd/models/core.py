@@ -179,12 +179,14 @@ class PairwiseConv(nn.Module):
         return k.reshape(*k.shape[:-4], self.d_out * self.nc_out,
                          to_order(self.degree_in) * self.nc_in)
 
-    def apply_fused(self, edge_feats, basis, x_gathered, max_chunk_bytes=1 << 28):
+    def apply_fused(self, edge_feats, basis, x_gathered, max_chunk_bytes=2 << 30):
         """out[..., mo, 2do+1] = sum_{mi,i,f} R[...,mo,mi,f] B[...,o,i,f] x[...,mi,i]
 
-        Streamed over edge chunks so the (mo*mi*F) radial activation never
-        exceeds ~max_chunk_bytes. This is the eager fallback mirror of the
-        fused HIP kernel.
+        Structure (mirrors the fused HIP kernel):
+          1. u = B.x once per pair — no channel^2 term, small (E*mi*F*O);
+          2. the radial output R (the E*mo*mi*F memory hog) is produced and
+             consumed chunk-by-chunk and, under grad, recomputed in backward
+             (activation checkpointing) instead of being stored.
         """
         O = self.d_out
         I = to_order(self.degree_in)
@@ -201,21 +203,17 @@ class PairwiseConv(nn.Module):
         bx = b.reshape(e_total, O, I, F_).to(x_gathered.dtype)
         xg = x_gathered.reshape(e_total, mi, I)
 
+        # u: (E, mi*F, O), (mi-major, f-minor) matching R's (mo, mi*F) layout
+        u = torch.einsum('eoif,eci->ecfo', bx, xg).reshape(e_total, mi * F_, O)
+
         elem_size = xg.element_size()
         per_edge = mo * mi * F_ * elem_size
-        chunk = max(1, min(e_total, max_chunk_bytes // max(per_edge, 1)))
+        chunk = max(256, min(e_total, max_chunk_bytes // max(per_edge, 1)))
 
-        def run_chunk(ef_c, bx_c, xg_c):
-            r = self.rp(ef_c).to(xg_c.dtype)                     # (E, mo, mi, F)
-            u = torch.einsum('eoif,eci->eocf', bx_c, xg_c)       # (E, O, mi, F)
-            return torch.matmul(
-                r.reshape(-1, mo, mi * F_),
-                u.permute(0, 2, 3, 1).reshape(-1, mi * F_, O))   # (E, mo, O)
+        def run_chunk(ef_c, u_c):
+            r = self.rp(ef_c).to(u_c.dtype).reshape(-1, mo, mi * F_)
+            return torch.bmm(r, u_c)                             # (E, mo, O)
 
-        # The per-chunk radial activation (E_chunk * mo * mi * F) is the
-        # memory hog; when training a large config, recompute it in backward
-        # (activation checkpointing) instead of keeping every chunk alive —
-        # the same recompute-R strategy the fused HIP backward uses.
         use_ckpt = torch.is_grad_enabled() and (e_total * per_edge > (1 << 26))
 
         outs = []
@@ -223,9 +221,9 @@ class PairwiseConv(nn.Module):
             e = slice(s, s + chunk)
             if use_ckpt:
                 out = torch.utils.checkpoint.checkpoint(
-                    run_chunk, ef[e], bx[e], xg[e], use_reentrant=False)
+                    run_chunk, ef[e], u[e], use_reentrant=False)
             else:
-                out = run_chunk(ef[e], bx[e], xg[e])
+                out = run_chunk(ef[e], u[e])
             outs.append(out)
         out = torch.cat(outs, dim=0) if len(outs) > 1 else outs[0]
         return out.view(*lead, mo, O)
