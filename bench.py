@@ -107,8 +107,15 @@ def main():
         if use_cuda:
             torch.cuda.synchronize()
 
-    for _ in range(args.warmup):
+    import sys
+    for i in range(args.warmup):
+        t = time.perf_counter()
         train_step()
+        if use_cuda:
+            torch.cuda.synchronize()
+        if rank == 0:
+            print(f'[bench] warmup {i}: {time.perf_counter() - t:.2f}s',
+                  file=sys.stderr, flush=True)
 
     barrier_sync()
     t0 = time.perf_counter()
