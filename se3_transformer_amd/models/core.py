@@ -15,6 +15,7 @@ distributions match. The COMPUTE is restructured MI355X-first:
 """
 from __future__ import annotations
 
+import os
 from math import sqrt
 
 import torch
@@ -202,6 +203,21 @@ class PairwiseConv(nn.Module):
         ef = edge_feats.reshape(e_total, edge_feats.shape[-1])
         bx = b.reshape(e_total, O, I, F_).to(x_gathered.dtype)
         xg = x_gathered.reshape(e_total, mi, I)
+
+        # MI355X fused HIP path: bf16 compute on CUDA(ROCm) devices
+        from ..ops import fused as _fused
+        want_bf16 = (torch.is_autocast_enabled()
+                     or xg.dtype == torch.bfloat16
+                     or os.environ.get('SE3_FORCE_FUSED') == '1')
+        if (xg.is_cuda and want_bf16
+                and _fused.fused_shapes_ok(mo, mi * F_, O, self.rp.mid_dim)):
+            _fused.require_ext()
+            h = self.rp.hidden(ef)                           # (E, 128)
+            # u_t[(mi,f), o, e] = sum_i B[e,o,i,f] x[e,mi,i]
+            u_t = torch.einsum('eoif,eci->cfoe', bx, xg).reshape(mi * F_, O, e_total)
+            w6 = self.rp.net[6]
+            out = _fused.fused_pairconv(h, w6.weight, w6.bias, u_t, mo)
+            return out.view(*lead, mo, O).to(xg.dtype)
 
         # u: (E, mi*F, O), (mi-major, f-minor) matching R's (mo, mi*F) layout
         u = torch.einsum('eoif,eci->ecfo', bx, xg).reshape(e_total, mi * F_, O)

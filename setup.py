@@ -1,0 +1,28 @@
+"""In-tree build of the MI355X (gfx950) HIP extensions.
+
+    PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+
+Produces se3_transformer_amd/_C*.so next to the package sources so the
+snapshot shipped to GPU boxes carries the built extension.
+"""
+import os
+
+os.environ.setdefault('PYTORCH_ROCM_ARCH', 'gfx950')
+
+from setuptools import setup
+from torch.utils.cpp_extension import BuildExtension, CUDAExtension
+
+setup(
+    name='se3_transformer_amd_ext',
+    ext_modules=[
+        CUDAExtension(
+            name='se3_transformer_amd._C',
+            sources=['se3_transformer_amd/csrc/pairconv.hip'],
+            extra_compile_args={
+                'cxx': ['-O3'],
+                'nvcc': ['-O3', '--offload-arch=gfx950'],
+            },
+        ),
+    ],
+    cmdclass={'build_ext': BuildExtension.with_options(no_python_abi_suffix=False)},
+)
