@@ -251,7 +251,17 @@ void pairconv_fwd(torch::Tensor H, torch::Tensor W, torch::Tensor Ut,
     TORCH_CHECK(err == hipSuccess, "pairconv_fwd launch failed: ", hipGetErrorString(err));
 }
 
+void pairconv_bwd_dh(torch::Tensor G, torch::Tensor Ut, torch::Tensor Wt,
+                     torch::Tensor dH, int64_t mo_);
+void pairconv_bwd_dw(torch::Tensor G, torch::Tensor Ut, torch::Tensor Ht,
+                     torch::Tensor dW, int64_t mo_);
+void pairconv_bwd_du(torch::Tensor H, torch::Tensor W, torch::Tensor bias,
+                     torch::Tensor G, torch::Tensor dU, int64_t mo_);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("pairconv_fwd", &pairconv_fwd,
           "fused radial-GEMM + basis contraction forward (MI355X)");
+    m.def("pairconv_bwd_dh", &pairconv_bwd_dh, "dH backward");
+    m.def("pairconv_bwd_dw", &pairconv_bwd_dw, "dW backward");
+    m.def("pairconv_bwd_du", &pairconv_bwd_du, "dU backward");
 }

@@ -1,0 +1,438 @@
+// Backward kernels for the fused pairwise convolution (see pairconv.hip).
+//
+//   R[e,n]  = H[e,:] . W[n,:] + bias[n],   n = (mo, c),  c = (mi,f) "urow"
+//   out[e,mo,o] = sum_c R[e,(mo,c)] * u[c,o,e]
+//
+// Gradients (g = dL/dout, (E,mo,O) bf16):
+//   dR[e,n]   = sum_o g[e,mo,o] * u[c,o,e]          (VALU, recomputed on the fly)
+//   dH[e,k]   = sum_n dR[e,n] * W[n,k]              (kernel B1, MFMA, needs W^T)
+//   dW[n,k]   = sum_e dR[e,n] * H[e,k]              (kernel B2, MFMA, needs H^T)
+//   db[n]     = sum_e dR[e,n]                       (folded into B2)
+//   du[c,o,e] = sum_mo R[e,(mo,c)] * g[e,mo,o]      (kernel B3, recomputes R)
+//
+// dR / R are never written to global memory anywhere.
+
+#include <torch/extension.h>
+#include <hip/hip_runtime.h>
+#include <ATen/hip/HIPContext.h>
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+#define KDIM 128
+#define NT 512
+
+__device__ __forceinline__ float b2f(__bf16 x) { return (float)x; }
+
+// ---------------------------------------------------------------------------
+// B1: dH[e,k] = sum_n dR[e,n] W[n,k]
+// block: 64 e x 128 k, loops (mo-block 8) x (urow-chunk 32) over all n.
+// Wt is W transposed: (128, mo*miF) bf16, so the MFMA B-operand
+// (8 consecutive n at fixed k) is a contiguous 16B load.
+// ---------------------------------------------------------------------------
+template <int O>
+__global__ void __launch_bounds__(NT)
+pairconv_bwd_dh_kernel(const __bf16* __restrict__ G,   // (E, mo, O) bf16
+                       const __bf16* __restrict__ Ut,  // (miF, O, E) bf16
+                       const __bf16* __restrict__ Wt,  // (128, mo*miF) bf16
+                       float* __restrict__ dH,         // (E, 128) f32
+                       int E, int mo, int miF) {
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    __bf16* dr_lds = reinterpret_cast<__bf16*>(smem);                 // [64e][256n] 32 KiB
+    __bf16* u_lds = reinterpret_cast<__bf16*>(smem + 32768);          // [32][O][64]
+    __bf16* g_lds = reinterpret_cast<__bf16*>(smem + 32768 + 32 * O * 64 * 2); // [8mo][O][64e]
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6;
+    const int l15 = lane & 15;
+    const int l4 = lane >> 4;
+    const int we = wid >> 1;          // 0..3: e-group of 16
+    const int wk = wid & 1;           // 0..1: k-group of 64
+    const int e0 = blockIdx.x * 64;
+    const size_t N = (size_t)mo * miF;
+
+    f32x4 acc[4];                      // 16 e x 64 k per wave
+#pragma unroll
+    for (int i = 0; i < 4; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+
+    const int nmo = mo / 8, nuc = miF / 32;
+    for (int mb = 0; mb < nmo; ++mb) {
+        // stage g tile [8][O][64]
+        for (int i = tid; i < 8 * O * 64; i += NT) {
+            int e = i & 63, rest = i >> 6;   // rest = m*O+o
+            int m = rest / O, o = rest % O;
+            g_lds[i] = (e0 + e < E) ? G[((size_t)(e0 + e) * mo + mb * 8 + m) * O + o]
+                                    : (__bf16)0.f;
+        }
+        for (int cb = 0; cb < nuc; ++cb) {
+            // stage u chunk [32][O][64]
+            __syncthreads();
+            for (int i = tid; i < (32 * O * 64) / 8; i += NT) {
+                int ro = i >> 3, eu = (i & 7) * 8;
+                const __bf16* src = Ut + ((size_t)(cb * 32 + ro / O) * O + (ro % O)) * E + e0 + eu;
+                bf16x8 v;
+                if (e0 + eu + 8 <= E) v = *reinterpret_cast<const bf16x8*>(src);
+                else { for (int j = 0; j < 8; ++j) v[j] = (e0 + eu + j < E) ? src[j] : (__bf16)0.f; }
+                *reinterpret_cast<bf16x8*>(u_lds + (size_t)ro * 64 + eu) = v;
+            }
+            __syncthreads();
+            // cooperative dR tile: [64e][256n], each thread 32 elements
+            for (int i = tid; i < 64 * 256; i += NT) {
+                int n = i & 255, e = i >> 8;
+                int m = n >> 5, c = n & 31;
+                float s = 0.f;
+#pragma unroll
+                for (int o = 0; o < O; ++o)
+                    s = fmaf(b2f(g_lds[(m * O + o) * 64 + e]),
+                             b2f(u_lds[(c * O + o) * 64 + e]), s);
+                // swizzled: 16B slot (n>>3) xor'd with (e&15) within the row
+                reinterpret_cast<__bf16*>(reinterpret_cast<char*>(dr_lds)
+                    + e * 512 + ((((n >> 3) ^ (e & 15)) << 4)))[n & 7] = (__bf16)s;
+            }
+            __syncthreads();
+            // MFMA: dH_tile += dR(64e x 256n) @ W(256n x 128k)
+            const size_t nbase = (size_t)(mb * 8) * miF + cb * 32;
+#pragma unroll
+            for (int ns = 0; ns < 8; ++ns) {      // 8 n-steps of 32
+                // A: dR rows e = we*16+l15, n = ns*32 + l4*8.. (mo-major inside tile)
+                int ntile = ns * 32 + l4 * 8;     // 0..255 (this is (m*32+c) packed)
+                int e_row = we * 16 + l15;
+                bf16x8 a = *reinterpret_cast<const bf16x8*>(
+                    reinterpret_cast<char*>(dr_lds) + e_row * 512
+                    + ((((ntile >> 3) ^ (e_row & 15)) << 4)));
+                // careful: dR tile n index = m*32+c, global n = (mb*8+m)*miF + cb*32 + c
+#pragma unroll
+                for (int kf = 0; kf < 4; ++kf) {
+                    int k = wk * 64 + kf * 16 + l15;
+                    // B[n][k] = Wt[k][n]: 8 consecutive n... but global n is NOT
+                    // contiguous across m boundaries; ns*32+l4*8 stays within one m
+                    // (32-aligned chunks of 32), so c-run of 8 is contiguous.
+                    int m = ntile >> 5, c = ntile & 31;
+                    const __bf16* src = Wt + (size_t)k * N + nbase + (size_t)m * miF + c;
+                    bf16x8 b = *reinterpret_cast<const bf16x8*>(src);
+                    acc[kf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(b, a, acc[kf], 0, 0, 0);
+                }
+            }
+        }
+    }
+    // D layout from mfma(b, a): D[row=k-ish?]; operands: A-arg=b (k rows), B-arg=a (e cols)
+    // mfma(X, Y, acc): D[i][j] = sum_k X[i][k(contraction)] ... X supplies M rows.
+    // Here X = b (Wt fragment: rows k, contraction n), Y = a (dR: contraction n, cols e)
+    // => D rows = k, cols = e: row = l4*4+reg (k), col = l15 (e).
+    {
+        for (int kf = 0; kf < 4; ++kf) {
+#pragma unroll
+            for (int reg = 0; reg < 4; ++reg) {
+                int k = wk * 64 + kf * 16 + l4 * 4 + reg;
+                int e = e0 + we * 16 + l15;
+                if (e < E) dH[(size_t)e * KDIM + k] = acc[kf][reg];
+            }
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
+// B2: dW[n,k] = sum_e dR[e,n] H[e,k];  db[n] = sum_e dR[e,n]
+// block: n-tile 128 (4 mo x 32 urow) x 128 k, loops e in chunks of 32.
+// Ht is H transposed: (128, E) bf16.
+// ---------------------------------------------------------------------------
+template <int O>
+__global__ void __launch_bounds__(NT)
+pairconv_bwd_dw_kernel(const __bf16* __restrict__ G,   // (E, mo, O)
+                       const __bf16* __restrict__ Ut,  // (miF, O, E)
+                       const __bf16* __restrict__ Ht,  // (128, E)
+                       float* __restrict__ dW,         // (mo*miF, 128) f32
+                       int E, int mo, int miF) {
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    __bf16* dr_lds = reinterpret_cast<__bf16*>(smem);                  // [128n][32e] 8 KiB
+    __bf16* h_lds = reinterpret_cast<__bf16*>(smem + 8192);            // [128k][32e] 8 KiB
+    __bf16* u_lds = reinterpret_cast<__bf16*>(smem + 16384);           // [32][O][32]
+    __bf16* g_lds = reinterpret_cast<__bf16*>(smem + 16384 + 32 * O * 32 * 2); // [4][O][32]
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6;
+    const int l15 = lane & 15;
+    const int l4 = lane >> 4;
+    const int wn = wid >> 1;          // 0..3: n-group of 32
+    const int wk = wid & 1;           // 0..1: k-group of 64
+    // n-tile: blockIdx.x over N/128; tile covers mo-block of 4, urow-chunk of 32
+    const int mb = blockIdx.x / (miF / 32);     // mo-block (4 mo each)
+    const int cb = blockIdx.x % (miF / 32);     // urow chunk
+
+    f32x4 acc[2][4];                  // wave: 32 n x 64 k
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+    const int nec = (E + 31) / 32;
+    for (int ec = 0; ec < nec; ++ec) {
+        const int e0 = ec * 32;
+        __syncthreads();
+        // stage u chunk [32c][O][32e] and g tile [4m][O][32e]
+        for (int i = tid; i < 32 * O * 32; i += NT) {
+            int e = i & 31, ro = i >> 5;  // c*O+o
+            u_lds[i] = (e0 + e < E)
+                ? Ut[((size_t)(cb * 32 + ro / O) * O + (ro % O)) * E + e0 + e]
+                : (__bf16)0.f;
+        }
+        for (int i = tid; i < 4 * O * 32; i += NT) {
+            int e = i & 31, rest = i >> 5;
+            int m = rest / O, o = rest % O;
+            g_lds[i] = (e0 + e < E)
+                ? G[((size_t)(e0 + e) * mo + mb * 4 + m) * O + o]
+                : (__bf16)0.f;
+        }
+        // stage H^T chunk [128k][32e]
+        for (int i = tid; i < (128 * 32) / 8; i += NT) {
+            int k = i >> 2, eu = (i & 3) * 8;
+            const __bf16* src = Ht + (size_t)k * E + e0 + eu;
+            bf16x8 v;
+            if (e0 + eu + 8 <= E) v = *reinterpret_cast<const bf16x8*>(src);
+            else { for (int j = 0; j < 8; ++j) v[j] = (e0 + eu + j < E) ? src[j] : (__bf16)0.f; }
+            *reinterpret_cast<bf16x8*>(h_lds + (size_t)k * 32 + eu) = v;
+        }
+        __syncthreads();
+        // cooperative dR^T tile [128n][32e] + db accumulation
+        for (int i = tid; i < 128 * 32; i += NT) {
+            int e = i & 31, n = i >> 5;
+            int m = n >> 5, c = n & 31;
+            float s = 0.f;
+#pragma unroll
+            for (int o = 0; o < O; ++o)
+                s = fmaf(b2f(g_lds[(m * O + o) * 32 + e]),
+                         b2f(u_lds[(c * O + o) * 32 + e]), s);
+            dr_lds[(size_t)n * 32 + e] = (__bf16)s;
+        }
+        __syncthreads();
+        // MFMA: dW_tile += dR^T(128n x 32e) @ H(32e x 128k)
+#pragma unroll
+        for (int es = 0; es < 1; ++es) {
+#pragma unroll
+            for (int nf = 0; nf < 2; ++nf) {
+                bf16x8 a = *reinterpret_cast<const bf16x8*>(
+                    dr_lds + (size_t)(wn * 32 + nf * 16 + l15) * 32 + l4 * 8);
+#pragma unroll
+                for (int kf = 0; kf < 4; ++kf) {
+                    bf16x8 b = *reinterpret_cast<const bf16x8*>(
+                        h_lds + (size_t)(wk * 64 + kf * 16 + l15) * 32 + l4 * 8);
+                    acc[nf][kf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nf][kf], 0, 0, 0);
+                }
+            }
+        }
+    }
+    __syncthreads();
+    // write dW: D rows = n ((l4*4+reg within frag)), cols = k (l15)
+    const size_t nbase = (size_t)(mb * 4) * miF + cb * 32;
+#pragma unroll
+    for (int nf = 0; nf < 2; ++nf) {
+#pragma unroll
+        for (int kf = 0; kf < 4; ++kf) {
+#pragma unroll
+            for (int reg = 0; reg < 4; ++reg) {
+                int ntile = wn * 32 + nf * 16 + l4 * 4 + reg;
+                int m = ntile >> 5, c = ntile & 31;
+                int k = wk * 64 + kf * 16 + l15;
+                dW[(nbase + (size_t)m * miF + c) * KDIM + k] = acc[nf][kf][reg];
+            }
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
+// B3: du[c,o,e] = sum_mo R[e,(mo,c)] g[e,mo,o],  R = H @ W^T + bias
+// block: 64 e x 32 urow, loops mo in blocks of 8; R tile recomputed by MFMA
+// exactly as the forward kernel, bounced through LDS, contracted vs g.
+// ---------------------------------------------------------------------------
+template <int O>
+__global__ void __launch_bounds__(NT)
+pairconv_bwd_du_kernel(const __bf16* __restrict__ H,   // (E,128)
+                       const __bf16* __restrict__ W,   // (mo*miF,128)
+                       const float* __restrict__ bias, // (mo*miF,)
+                       const __bf16* __restrict__ G,   // (E,mo,O)
+                       float* __restrict__ dU,         // (miF, O, E) f32
+                       int E, int mo, int miF) {
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    __bf16* h_lds = reinterpret_cast<__bf16*>(smem);                   // [64][128] swizzled 16K
+    __bf16* r_lds = reinterpret_cast<__bf16*>(smem + 16384);           // [256n][64e] 32K
+    __bf16* g_lds = reinterpret_cast<__bf16*>(smem + 49152);           // [8][O][64]
+    float* du_acc = reinterpret_cast<float*>(smem + 49152 + ((8 * O * 64 * 2 + 15) & ~15)); // [32][O][64]
+    float* bias_lds = reinterpret_cast<float*>(
+        smem + 49152 + ((8 * O * 64 * 2 + 15) & ~15) + 32 * O * 64 * 4);  // [256]
+
+    const int tid = threadIdx.x;
+    const int lane = tid & 63;
+    const int wid = tid >> 6;
+    const int l15 = lane & 15;
+    const int l4 = lane >> 4;
+    const int wm = wid >> 1;
+    const int we = wid & 1;
+    const int e0 = blockIdx.x * 64;
+    const int cb = blockIdx.y;        // urow chunk
+    const int uc0 = cb * 32;
+
+    // stage H tile swizzled (as forward)
+    for (int i = tid; i < (64 * KDIM) / 8; i += NT) {
+        int e = i >> 4, k16 = i & 15;
+        int dst = e * 256 + ((k16 * 16) ^ ((e & 15) << 4));
+        bf16x8 v;
+        if (e0 + e < E) v = *reinterpret_cast<const bf16x8*>(H + (size_t)(e0 + e) * KDIM + k16 * 8);
+        else v = bf16x8(0);
+        *reinterpret_cast<bf16x8*>(reinterpret_cast<char*>(h_lds) + dst) = v;
+    }
+    for (int i = tid; i < 32 * O * 64; i += NT) du_acc[i] = 0.f;
+    __syncthreads();
+
+    const int nmo = mo / 8;
+    for (int mb = 0; mb < nmo; ++mb) {
+        // stage g tile [8][O][64] and bias chunk [256]
+        for (int i = tid; i < 8 * O * 64; i += NT) {
+            int e = i & 63, rest = i >> 6;
+            int m = rest / O, o = rest % O;
+            g_lds[i] = (e0 + e < E)
+                ? G[((size_t)(e0 + e) * mo + mb * 8 + m) * O + o]
+                : (__bf16)0.f;
+        }
+        for (int i = tid; i < 256; i += NT) {
+            int m = i >> 5, c = i & 31;
+            bias_lds[i] = bias[(size_t)(mb * 8 + m) * miF + uc0 + c];
+        }
+        __syncthreads();
+        // MFMA R tile: (256n x 64e) like forward
+        f32x4 acc[4][2];
+#pragma unroll
+        for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+            for (int ef = 0; ef < 2; ++ef) acc[mf][ef] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int kit = 0; kit < 4; ++kit) {
+            const int k0 = kit * 32 + l4 * 8;
+            bf16x8 a[4], b[2];
+#pragma unroll
+            for (int mf = 0; mf < 4; ++mf) {
+                int r = wm * 64 + mf * 16 + l15;
+                size_t n = (size_t)(mb * 8 + (r >> 5)) * miF + uc0 + (r & 31);
+                a[mf] = *reinterpret_cast<const bf16x8*>(W + n * KDIM + k0);
+            }
+#pragma unroll
+            for (int ef = 0; ef < 2; ++ef) {
+                int e = we * 32 + ef * 16 + l15;
+                int byte = e * 256 + ((k0 * 2) ^ ((e & 15) << 4));
+                b[ef] = *reinterpret_cast<const bf16x8*>(reinterpret_cast<char*>(h_lds) + byte);
+            }
+#pragma unroll
+            for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+                for (int ef = 0; ef < 2; ++ef)
+                    acc[mf][ef] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[mf], b[ef], acc[mf][ef], 0, 0, 0);
+        }
+        // bounce R (+bias) to LDS [256n][64e]
+#pragma unroll
+        for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+            for (int ef = 0; ef < 2; ++ef)
+#pragma unroll
+                for (int reg = 0; reg < 4; ++reg) {
+                    int r = wm * 64 + mf * 16 + l4 * 4 + reg;
+                    int e = we * 32 + ef * 16 + l15;
+                    r_lds[(size_t)r * 64 + e] = (__bf16)(acc[mf][ef][reg] + bias_lds[r]);
+                }
+        __syncthreads();
+        // contraction: du_acc[c][o][e] += sum_m R[(m,c)][e] * g[m][o][e]
+        // partition (c,e) across all 512 threads: 32*64 = 2048 cells, 4 per thread
+        for (int i = tid; i < 32 * 64; i += NT) {
+            int e = i & 63, c = i >> 6;
+#pragma unroll
+            for (int o = 0; o < O; ++o) {
+                float s = du_acc[(c * O + o) * 64 + e];
+#pragma unroll
+                for (int m = 0; m < 8; ++m)
+                    s = fmaf(b2f(r_lds[(size_t)(m * 32 + c) * 64 + e]),
+                             b2f(g_lds[(m * O + o) * 64 + e]), s);
+                du_acc[(c * O + o) * 64 + e] = s;
+            }
+        }
+        __syncthreads();
+    }
+    // write du chunk
+    for (int i = tid; i < 32 * O * 64; i += NT) {
+        int e = i & 63, ro = i >> 6;  // c*O+o
+        if (e0 + e < E)
+            dU[((size_t)(uc0 + ro / O) * O + (ro % O)) * E + e0 + e] = du_acc[i];
+    }
+}
+
+// ---------------------------------------------------------------------------
+// launchers
+// ---------------------------------------------------------------------------
+#define DISPATCH_O(O, ...)                                         \
+    switch (O) {                                                   \
+        case 1: { constexpr int kO = 1; __VA_ARGS__; break; }      \
+        case 3: { constexpr int kO = 3; __VA_ARGS__; break; }      \
+        case 5: { constexpr int kO = 5; __VA_ARGS__; break; }      \
+        case 7: { constexpr int kO = 7; __VA_ARGS__; break; }      \
+        default: TORCH_CHECK(false, "unsupported O ", O);          \
+    }
+
+void pairconv_bwd_dh(torch::Tensor G, torch::Tensor Ut, torch::Tensor Wt,
+                     torch::Tensor dH, int64_t mo_) {
+    int E = dH.size(0), mo = (int)mo_, miF = Ut.size(0), O = Ut.size(1);
+    TORCH_CHECK(G.is_contiguous() && Ut.is_contiguous() && Wt.is_contiguous() && dH.is_contiguous());
+    TORCH_CHECK(Wt.size(0) == KDIM && Wt.size(1) == (int64_t)mo * miF);
+    auto stream = at::cuda::getCurrentHIPStream();
+    dim3 grid((E + 63) / 64);
+    DISPATCH_O(O, {
+        size_t lds = 32768 + (size_t)32 * kO * 64 * 2 + (size_t)8 * kO * 64 * 2;
+        hipLaunchKernelGGL(HIP_KERNEL_NAME(pairconv_bwd_dh_kernel<kO>), grid, dim3(NT), lds, stream,
+                           reinterpret_cast<const __bf16*>(G.data_ptr()),
+                           reinterpret_cast<const __bf16*>(Ut.data_ptr()),
+                           reinterpret_cast<const __bf16*>(Wt.data_ptr()),
+                           dH.data_ptr<float>(), E, mo, miF);
+    });
+    hipError_t err = hipGetLastError();
+    TORCH_CHECK(err == hipSuccess, "bwd_dh: ", hipGetErrorString(err));
+}
+
+void pairconv_bwd_dw(torch::Tensor G, torch::Tensor Ut, torch::Tensor Ht,
+                     torch::Tensor dW, int64_t mo_) {
+    int E = Ht.size(1), mo = (int)mo_, miF = Ut.size(0), O = Ut.size(1);
+    TORCH_CHECK(G.is_contiguous() && Ut.is_contiguous() && Ht.is_contiguous() &&
+                dW.is_contiguous());
+    TORCH_CHECK(mo % 4 == 0 && miF % 32 == 0);
+    auto stream = at::cuda::getCurrentHIPStream();
+    dim3 grid((mo / 4) * (miF / 32));
+    DISPATCH_O(O, {
+        size_t lds = 16384 + (size_t)32 * kO * 32 * 2 + (size_t)4 * kO * 32 * 2;
+        hipLaunchKernelGGL(HIP_KERNEL_NAME(pairconv_bwd_dw_kernel<kO>), grid, dim3(NT), lds, stream,
+                           reinterpret_cast<const __bf16*>(G.data_ptr()),
+                           reinterpret_cast<const __bf16*>(Ut.data_ptr()),
+                           reinterpret_cast<const __bf16*>(Ht.data_ptr()),
+                           dW.data_ptr<float>(), E, mo, miF);
+    });
+    hipError_t err = hipGetLastError();
+    TORCH_CHECK(err == hipSuccess, "bwd_dw: ", hipGetErrorString(err));
+}
+
+void pairconv_bwd_du(torch::Tensor H, torch::Tensor W, torch::Tensor bias,
+                     torch::Tensor G, torch::Tensor dU, int64_t mo_) {
+    int E = H.size(0), mo = (int)mo_, miF = dU.size(0), O = dU.size(1);
+    TORCH_CHECK(H.is_contiguous() && W.is_contiguous() && G.is_contiguous() &&
+                bias.is_contiguous() && dU.is_contiguous());
+    TORCH_CHECK(bias.dtype() == torch::kFloat32);
+    auto stream = at::cuda::getCurrentHIPStream();
+    dim3 grid((E + 63) / 64, miF / 32);
+    DISPATCH_O(O, {
+        size_t lds = 49152 + (size_t)((8 * kO * 64 * 2 + 15) & ~15) +
+                     (size_t)32 * kO * 64 * 4 + 256 * 4;
+        hipLaunchKernelGGL(HIP_KERNEL_NAME(pairconv_bwd_du_kernel<kO>), grid, dim3(NT), lds, stream,
+                           reinterpret_cast<const __bf16*>(H.data_ptr()),
+                           reinterpret_cast<const __bf16*>(W.data_ptr()),
+                           bias.data_ptr<float>(),
+                           reinterpret_cast<const __bf16*>(G.data_ptr()),
+                           dU.data_ptr<float>(), E, mo, miF);
+    });
+    hipError_t err = hipGetLastError();
+    TORCH_CHECK(err == hipSuccess, "bwd_du: ", hipGetErrorString(err));
+}

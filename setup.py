@@ -17,7 +17,8 @@ setup(
     ext_modules=[
         CUDAExtension(
             name='se3_transformer_amd._C',
-            sources=['se3_transformer_amd/csrc/pairconv.hip'],
+            sources=['se3_transformer_amd/csrc/pairconv.hip',
+                     'se3_transformer_amd/csrc/pairconv_bwd.hip'],
             extra_compile_args={
                 'cxx': ['-O3'],
                 'nvcc': ['-O3', '--offload-arch=gfx950'],

@@ -72,6 +72,8 @@ def test_pairconv_kernel_backward_parity():
     ref = pc.apply_fused(ef, {(di, do): b}, xg0).float()
     ref.pow(2).mean().backward()
     gw_ref = pc.rp.net[6].weight.grad.clone()
+    gb_ref = pc.rp.net[6].bias.grad.clone()
+    g0_ref = pc.rp.net[0].weight.grad.clone()
     gx_ref = xg0.grad.clone()
     pc.zero_grad()
 
@@ -82,11 +84,56 @@ def test_pairconv_kernel_backward_parity():
     finally:
         del os.environ['SE3_FORCE_FUSED']
     gw = pc.rp.net[6].weight.grad.clone()
+    gb = pc.rp.net[6].bias.grad.clone()
+    g0 = pc.rp.net[0].weight.grad.clone()
     gx = xg1.grad.clone()
 
     assert _rel_err(out, ref) < 5e-2
-    assert _rel_err(gx, gx_ref) < 8e-2
-    assert _rel_err(gw.float(), gw_ref.float()) < 8e-2
+    assert _rel_err(gx, gx_ref) < 8e-2, 'du path'
+    assert _rel_err(gw.float(), gw_ref.float()) < 8e-2, 'dW path'
+    assert _rel_err(gb.float(), gb_ref.float()) < 8e-2, 'db path'
+    assert _rel_err(g0.float(), g0_ref.float()) < 8e-2, 'dH path (trunk)'
+
+
+@needs_gpu
+@pytest.mark.parametrize('di,do', [(0, 0), (3, 3)])
+def test_hip_bwd_vs_torch_bwd(di, do):
+    """The three HIP backward kernels vs the torch chunked backward of the
+    same fused Function (identical bf16 forward, so tight tolerance)."""
+    from se3_transformer_amd.models.core import PairwiseConv
+
+    torch.manual_seed(2)
+    device = torch.device('cuda')
+    E, mi, mo = 777, 64, 32
+    pc = PairwiseConv(di, mi, do, mo, edge_dim=1).to(device)
+    F_, O, I = pc.num_freq, 2 * do + 1, 2 * di + 1
+    ef = torch.randn(E, 2, device=device)
+    b = torch.randn(E, O, I, F_, device=device)
+
+    grads = {}
+    for mode in ('hip', 'torch'):
+        xg = torch.randn(E, mi, I, device=device,
+                         generator=torch.Generator(device).manual_seed(7),
+                         requires_grad=True)
+        pc.zero_grad()
+        os.environ['SE3_FORCE_FUSED'] = '1'
+        if mode == 'torch':
+            os.environ['SE3_TORCH_BWD'] = '1'
+        try:
+            out = pc.apply_fused(ef, {(di, do): b}, xg).float()
+            out.pow(2).mean().backward()
+        finally:
+            del os.environ['SE3_FORCE_FUSED']
+            os.environ.pop('SE3_TORCH_BWD', None)
+        grads[mode] = {
+            'x': xg.grad.clone(),
+            'w6': pc.rp.net[6].weight.grad.clone(),
+            'b6': pc.rp.net[6].bias.grad.clone(),
+            'w0': pc.rp.net[0].weight.grad.clone(),
+        }
+    for k in grads['hip']:
+        err = _rel_err(grads['hip'][k].float(), grads['torch'][k].float())
+        assert err < 2e-2, f'{k}: {err}'
 
 
 @needs_gpu
