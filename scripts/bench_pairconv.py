@@ -1,0 +1,82 @@
+"""Microbenchmark the fused pairconv kernels at headline shapes.
+
+Usage (GPU box):  python scripts/bench_pairconv.py [--pair 3,3]
+Prints per-kernel ms and effective TFLOP/s for fwd, bwd_dh, bwd_dw, bwd_du.
+"""
+import argparse
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+from se3_transformer_amd import _C
+
+
+def timeit(fn, iters=10, warmup=3):
+    for _ in range(warmup):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / iters * 1000
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument('--pair', default='3,3')
+    p.add_argument('--E', type=int, default=9216)
+    p.add_argument('--ch', type=int, default=512)
+    args = p.parse_args()
+    di, do = map(int, args.pair.split(','))
+    F = 2 * min(di, do) + 1
+    O = 2 * do + 1
+    mi = mo = args.ch
+    miF = mi * F
+    E, K = args.E, 128
+    N = mo * miF
+    dev = 'cuda'
+    g = torch.Generator(device=dev).manual_seed(0)
+
+    H = torch.randn(E, K, generator=g, device=dev).to(torch.bfloat16)
+    W = (torch.randn(N, K, generator=g, device=dev) / K**0.5).to(torch.bfloat16)
+    Ut = torch.randn(miF, O, E, generator=g, device=dev).to(torch.bfloat16)
+    out = torch.zeros(E, mo, O, device=dev)
+    gt = torch.randn(mo, O, E, generator=g, device=dev).to(torch.bfloat16)
+    bias = torch.zeros(N, device=dev)
+
+    gemm_fl = 2.0 * E * N * K
+    epi_fl = 2.0 * E * N * O
+
+    ms = timeit(lambda: _C.pairconv_fwd(H, W, Ut, out, mo))
+    print(f'fwd    ({di},{do}): {ms:8.3f} ms  {gemm_fl/ms/1e9:7.1f} TF/s (gemm) '
+          f'{(gemm_fl+epi_fl)/ms/1e9:7.1f} TF/s (total)')
+
+    Wt = W.t().contiguous()
+    dH = torch.zeros(E, K, device=dev)
+    ms = timeit(lambda: _C.pairconv_bwd_dh(gt, Ut, Wt, dH, mo))
+    print(f'bwd_dh ({di},{do}): {ms:8.3f} ms  {gemm_fl/ms/1e9:7.1f} TF/s')
+
+    Ht = H.t().contiguous()
+    dW = torch.empty(N, K, device=dev)
+    ms = timeit(lambda: _C.pairconv_bwd_dw(gt, Ut, Ht, dW, mo))
+    print(f'bwd_dw ({di},{do}): {ms:8.3f} ms  {gemm_fl/ms/1e9:7.1f} TF/s')
+
+    dU = torch.empty(miF, O, E, device=dev)
+    ms = timeit(lambda: _C.pairconv_bwd_du(H, W, bias, gt, dU, mo))
+    print(f'bwd_du ({di},{do}): {ms:8.3f} ms  {gemm_fl/ms/1e9:7.1f} TF/s')
+
+    # library GEMM reference: the raw R = H @ W^T (what eager must do, without
+    # even the contraction), on a slab 1/8 of N to fit memory
+    Nr = N // 8
+    Wr = W[:Nr]
+    ms = timeit(lambda: H @ Wr.t(), iters=5)
+    print(f'hipblaslt H@W[{Nr}] : {ms:8.3f} ms  {2.0*E*Nr*K/ms/1e9:7.1f} TF/s'
+          f'  (writes R slab: {E*Nr*2/1e9:.2f} GB)')
+
+
+if __name__ == '__main__':
+    main()

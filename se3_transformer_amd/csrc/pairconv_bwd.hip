@@ -32,11 +32,11 @@ __device__ __forceinline__ float b2f(__bf16 x) { return (float)x; }
 // ---------------------------------------------------------------------------
 template <int O>
 __global__ void __launch_bounds__(NT)
-pairconv_bwd_dh_kernel(const __bf16* __restrict__ G,   // (E, mo, O) bf16
+pairconv_bwd_dh_kernel(const __bf16* __restrict__ Gt,  // (mo, O, E) bf16
                        const __bf16* __restrict__ Ut,  // (miF, O, E) bf16
                        const __bf16* __restrict__ Wt,  // (128, mo*miF) bf16
-                       float* __restrict__ dH,         // (E, 128) f32
-                       int E, int mo, int miF) {
+                       float* __restrict__ dH,         // (E, 128) f32 (zeroed)
+                       int E, int mo, int miF, int nsplit) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
     __bf16* dr_lds = reinterpret_cast<__bf16*>(smem);                 // [64e][256n] 32 KiB
     __bf16* u_lds = reinterpret_cast<__bf16*>(smem + 32768);          // [32][O][64]
@@ -57,13 +57,17 @@ pairconv_bwd_dh_kernel(const __bf16* __restrict__ G,   // (E, mo, O) bf16
     for (int i = 0; i < 4; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
 
     const int nmo = mo / 8, nuc = miF / 32;
-    for (int mb = 0; mb < nmo; ++mb) {
-        // stage g tile [8][O][64]
-        for (int i = tid; i < 8 * O * 64; i += NT) {
-            int e = i & 63, rest = i >> 6;   // rest = m*O+o
-            int m = rest / O, o = rest % O;
-            g_lds[i] = (e0 + e < E) ? G[((size_t)(e0 + e) * mo + mb * 8 + m) * O + o]
-                                    : (__bf16)0.f;
+    const int mb_lo = (nmo / nsplit) * blockIdx.y;
+    const int mb_hi = mb_lo + nmo / nsplit;
+    for (int mb = mb_lo; mb < mb_hi; ++mb) {
+        // stage g tile [8][O][64] from Gt (e-contiguous rows)
+        for (int i = tid; i < (8 * O * 64) / 8; i += NT) {
+            int ro = i >> 3, eu = (i & 7) * 8;   // ro = m*O+o
+            const __bf16* src = Gt + ((size_t)(mb * 8 + ro / O) * O + (ro % O)) * E + e0 + eu;
+            bf16x8 v;
+            if (e0 + eu + 8 <= E) v = *reinterpret_cast<const bf16x8*>(src);
+            else { for (int j = 0; j < 8; ++j) v[j] = (e0 + eu + j < E) ? src[j] : (__bf16)0.f; }
+            *reinterpret_cast<bf16x8*>(g_lds + (size_t)ro * 64 + eu) = v;
         }
         for (int cb = 0; cb < nuc; ++cb) {
             // stage u chunk [32][O][64]
@@ -126,7 +130,7 @@ pairconv_bwd_dh_kernel(const __bf16* __restrict__ G,   // (E, mo, O) bf16
             for (int reg = 0; reg < 4; ++reg) {
                 int k = wk * 64 + kf * 16 + l4 * 4 + reg;
                 int e = e0 + we * 16 + l15;
-                if (e < E) dH[(size_t)e * KDIM + k] = acc[kf][reg];
+                if (e < E) atomicAdd(&dH[(size_t)e * KDIM + k], acc[kf][reg]);
             }
         }
     }
@@ -139,7 +143,7 @@ pairconv_bwd_dh_kernel(const __bf16* __restrict__ G,   // (E, mo, O) bf16
 // ---------------------------------------------------------------------------
 template <int O>
 __global__ void __launch_bounds__(NT)
-pairconv_bwd_dw_kernel(const __bf16* __restrict__ G,   // (E, mo, O)
+pairconv_bwd_dw_kernel(const __bf16* __restrict__ Gt,  // (mo, O, E)
                        const __bf16* __restrict__ Ut,  // (miF, O, E)
                        const __bf16* __restrict__ Ht,  // (128, E)
                        float* __restrict__ dW,         // (mo*miF, 128) f32
@@ -171,19 +175,22 @@ pairconv_bwd_dw_kernel(const __bf16* __restrict__ G,   // (E, mo, O)
     for (int ec = 0; ec < nec; ++ec) {
         const int e0 = ec * 32;
         __syncthreads();
-        // stage u chunk [32c][O][32e] and g tile [4m][O][32e]
-        for (int i = tid; i < 32 * O * 32; i += NT) {
-            int e = i & 31, ro = i >> 5;  // c*O+o
-            u_lds[i] = (e0 + e < E)
-                ? Ut[((size_t)(cb * 32 + ro / O) * O + (ro % O)) * E + e0 + e]
-                : (__bf16)0.f;
+        // stage u chunk [32c][O][32e] and g tile [4m][O][32e] (16B units)
+        for (int i = tid; i < (32 * O * 32) / 8; i += NT) {
+            int ro = i >> 2, eu = (i & 3) * 8;
+            const __bf16* src = Ut + ((size_t)(cb * 32 + ro / O) * O + (ro % O)) * E + e0 + eu;
+            bf16x8 v;
+            if (e0 + eu + 8 <= E) v = *reinterpret_cast<const bf16x8*>(src);
+            else { for (int j = 0; j < 8; ++j) v[j] = (e0 + eu + j < E) ? src[j] : (__bf16)0.f; }
+            *reinterpret_cast<bf16x8*>(u_lds + (size_t)ro * 32 + eu) = v;
         }
-        for (int i = tid; i < 4 * O * 32; i += NT) {
-            int e = i & 31, rest = i >> 5;
-            int m = rest / O, o = rest % O;
-            g_lds[i] = (e0 + e < E)
-                ? G[((size_t)(e0 + e) * mo + mb * 4 + m) * O + o]
-                : (__bf16)0.f;
+        for (int i = tid; i < (4 * O * 32) / 8; i += NT) {
+            int ro = i >> 2, eu = (i & 3) * 8;
+            const __bf16* src = Gt + ((size_t)(mb * 4 + ro / O) * O + (ro % O)) * E + e0 + eu;
+            bf16x8 v;
+            if (e0 + eu + 8 <= E) v = *reinterpret_cast<const bf16x8*>(src);
+            else { for (int j = 0; j < 8; ++j) v[j] = (e0 + eu + j < E) ? src[j] : (__bf16)0.f; }
+            *reinterpret_cast<bf16x8*>(g_lds + (size_t)ro * 32 + eu) = v;
         }
         // stage H^T chunk [128k][32e]
         for (int i = tid; i < (128 * 32) / 8; i += NT) {
@@ -251,7 +258,7 @@ __global__ void __launch_bounds__(NT)
 pairconv_bwd_du_kernel(const __bf16* __restrict__ H,   // (E,128)
                        const __bf16* __restrict__ W,   // (mo*miF,128)
                        const float* __restrict__ bias, // (mo*miF,)
-                       const __bf16* __restrict__ G,   // (E,mo,O)
+                       const __bf16* __restrict__ Gt,  // (mo,O,E)
                        float* __restrict__ dU,         // (miF, O, E) f32
                        int E, int mo, int miF) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -287,13 +294,14 @@ pairconv_bwd_du_kernel(const __bf16* __restrict__ H,   // (E,128)
 
     const int nmo = mo / 8;
     for (int mb = 0; mb < nmo; ++mb) {
-        // stage g tile [8][O][64] and bias chunk [256]
-        for (int i = tid; i < 8 * O * 64; i += NT) {
-            int e = i & 63, rest = i >> 6;
-            int m = rest / O, o = rest % O;
-            g_lds[i] = (e0 + e < E)
-                ? G[((size_t)(e0 + e) * mo + mb * 8 + m) * O + o]
-                : (__bf16)0.f;
+        // stage g tile [8][O][64] (16B units) and bias chunk [256]
+        for (int i = tid; i < (8 * O * 64) / 8; i += NT) {
+            int ro = i >> 3, eu = (i & 7) * 8;
+            const __bf16* src = Gt + ((size_t)(mb * 8 + ro / O) * O + (ro % O)) * E + e0 + eu;
+            bf16x8 v;
+            if (e0 + eu + 8 <= E) v = *reinterpret_cast<const bf16x8*>(src);
+            else { for (int j = 0; j < 8; ++j) v[j] = (e0 + eu + j < E) ? src[j] : (__bf16)0.f; }
+            *reinterpret_cast<bf16x8*>(g_lds + (size_t)ro * 64 + eu) = v;
         }
         for (int i = tid; i < 256; i += NT) {
             int m = i >> 5, c = i & 31;
@@ -376,29 +384,34 @@ pairconv_bwd_du_kernel(const __bf16* __restrict__ H,   // (E,128)
         default: TORCH_CHECK(false, "unsupported O ", O);          \
     }
 
-void pairconv_bwd_dh(torch::Tensor G, torch::Tensor Ut, torch::Tensor Wt,
+void pairconv_bwd_dh(torch::Tensor Gt, torch::Tensor Ut, torch::Tensor Wt,
                      torch::Tensor dH, int64_t mo_) {
     int E = dH.size(0), mo = (int)mo_, miF = Ut.size(0), O = Ut.size(1);
-    TORCH_CHECK(G.is_contiguous() && Ut.is_contiguous() && Wt.is_contiguous() && dH.is_contiguous());
+    TORCH_CHECK(Gt.is_contiguous() && Ut.is_contiguous() && Wt.is_contiguous() && dH.is_contiguous());
     TORCH_CHECK(Wt.size(0) == KDIM && Wt.size(1) == (int64_t)mo * miF);
     auto stream = at::cuda::getCurrentHIPStream();
-    dim3 grid((E + 63) / 64);
+    int eblk = (E + 63) / 64;
+    int nmo = mo / 8;
+    int nsplit = 1;
+    while (eblk * nsplit * 2 <= 1024 && nsplit * 2 <= nmo && nmo % (nsplit * 2) == 0)
+        nsplit *= 2;
+    dim3 grid(eblk, nsplit);
     DISPATCH_O(O, {
         size_t lds = 32768 + (size_t)32 * kO * 64 * 2 + (size_t)8 * kO * 64 * 2;
         hipLaunchKernelGGL(HIP_KERNEL_NAME(pairconv_bwd_dh_kernel<kO>), grid, dim3(NT), lds, stream,
-                           reinterpret_cast<const __bf16*>(G.data_ptr()),
+                           reinterpret_cast<const __bf16*>(Gt.data_ptr()),
                            reinterpret_cast<const __bf16*>(Ut.data_ptr()),
                            reinterpret_cast<const __bf16*>(Wt.data_ptr()),
-                           dH.data_ptr<float>(), E, mo, miF);
+                           dH.data_ptr<float>(), E, mo, miF, nsplit);
     });
     hipError_t err = hipGetLastError();
     TORCH_CHECK(err == hipSuccess, "bwd_dh: ", hipGetErrorString(err));
 }
 
-void pairconv_bwd_dw(torch::Tensor G, torch::Tensor Ut, torch::Tensor Ht,
+void pairconv_bwd_dw(torch::Tensor Gt, torch::Tensor Ut, torch::Tensor Ht,
                      torch::Tensor dW, int64_t mo_) {
     int E = Ht.size(1), mo = (int)mo_, miF = Ut.size(0), O = Ut.size(1);
-    TORCH_CHECK(G.is_contiguous() && Ut.is_contiguous() && Ht.is_contiguous() &&
+    TORCH_CHECK(Gt.is_contiguous() && Ut.is_contiguous() && Ht.is_contiguous() &&
                 dW.is_contiguous());
     TORCH_CHECK(mo % 4 == 0 && miF % 32 == 0);
     auto stream = at::cuda::getCurrentHIPStream();
@@ -406,7 +419,7 @@ void pairconv_bwd_dw(torch::Tensor G, torch::Tensor Ut, torch::Tensor Ht,
     DISPATCH_O(O, {
         size_t lds = 16384 + (size_t)32 * kO * 32 * 2 + (size_t)4 * kO * 32 * 2;
         hipLaunchKernelGGL(HIP_KERNEL_NAME(pairconv_bwd_dw_kernel<kO>), grid, dim3(NT), lds, stream,
-                           reinterpret_cast<const __bf16*>(G.data_ptr()),
+                           reinterpret_cast<const __bf16*>(Gt.data_ptr()),
                            reinterpret_cast<const __bf16*>(Ut.data_ptr()),
                            reinterpret_cast<const __bf16*>(Ht.data_ptr()),
                            dW.data_ptr<float>(), E, mo, miF);
@@ -416,9 +429,9 @@ void pairconv_bwd_dw(torch::Tensor G, torch::Tensor Ut, torch::Tensor Ht,
 }
 
 void pairconv_bwd_du(torch::Tensor H, torch::Tensor W, torch::Tensor bias,
-                     torch::Tensor G, torch::Tensor dU, int64_t mo_) {
+                     torch::Tensor Gt, torch::Tensor dU, int64_t mo_) {
     int E = H.size(0), mo = (int)mo_, miF = dU.size(0), O = dU.size(1);
-    TORCH_CHECK(H.is_contiguous() && W.is_contiguous() && G.is_contiguous() &&
+    TORCH_CHECK(H.is_contiguous() && W.is_contiguous() && Gt.is_contiguous() &&
                 bias.is_contiguous() && dU.is_contiguous());
     TORCH_CHECK(bias.dtype() == torch::kFloat32);
     auto stream = at::cuda::getCurrentHIPStream();
@@ -430,7 +443,7 @@ void pairconv_bwd_du(torch::Tensor H, torch::Tensor W, torch::Tensor bias,
                            reinterpret_cast<const __bf16*>(H.data_ptr()),
                            reinterpret_cast<const __bf16*>(W.data_ptr()),
                            bias.data_ptr<float>(),
-                           reinterpret_cast<const __bf16*>(G.data_ptr()),
+                           reinterpret_cast<const __bf16*>(Gt.data_ptr()),
                            dU.data_ptr<float>(), E, mo, miF);
     });
     hipError_t err = hipGetLastError();

@@ -88,27 +88,28 @@ class _FusedPairConv(torch.autograd.Function):
         ext = _load_ext()
         if ext is not None and os.environ.get('SE3_TORCH_BWD') != '1':
             dH = dW = db = dUt = None
+            g_t = g16.permute(1, 2, 0).contiguous()    # (mo, O, E) bf16
             if need_H:
                 Wt = W16.t().contiguous()              # (128, N)
-                dH = torch.empty(E, K, dtype=torch.float32, device=H16.device)
-                ext.pairconv_bwd_dh(g16, Ut16, Wt, dH, mo)
+                dH = torch.zeros(E, K, dtype=torch.float32, device=H16.device)
+                ext.pairconv_bwd_dh(g_t, Ut16, Wt, dH, mo)
                 del Wt
             if need_W:
                 Ht = H16.t().contiguous()              # (128, E)
                 dW = torch.empty(mo * miF, K, dtype=torch.float32,
                                  device=H16.device)
-                ext.pairconv_bwd_dw(g16, Ut16, Ht, dW, mo)
+                ext.pairconv_bwd_dw(g_t, Ut16, Ht, dW, mo)
                 dW = dW.to(ctx.w_dtype)
                 del Ht
             if need_b:
-                # db[(m,c)] = sum_{e,o} g[e,m,o] u[c,o,e]  — one real GEMM
-                g_r = g16.permute(1, 0, 2).reshape(mo, E * O)
-                u_r = Ut16.permute(0, 2, 1).reshape(miF, E * O)
-                db = (g_r @ u_r.t()).reshape(mo * miF).to(ctx.b_dtype)
+                # db[(m,c)] = sum_{e,o} g[e,m,o] u[c,o,e] — one GEMM, both
+                # operands are free (mo,O*E)/(miF,O*E) views
+                db = (g_t.reshape(mo, O * E) @ Ut16.reshape(miF, O * E).t()) \
+                    .reshape(mo * miF).to(ctx.b_dtype)
             if need_u:
                 dUt = torch.empty(miF, O, E, dtype=torch.float32,
                                   device=H16.device)
-                ext.pairconv_bwd_du(H16, W16, b16.float().reshape(-1), g16,
+                ext.pairconv_bwd_du(H16, W16, b16.float().reshape(-1), g_t,
                                     dUt, mo)
             return dH, dW, db, dUt, None
 
