@@ -51,13 +51,15 @@ def main():
     gemm_fl = 2.0 * E * N * K
     epi_fl = 2.0 * E * N * O
 
-    ms = timeit(lambda: _C.pairconv_fwd(H, W, Ut, out, mo))
+    from se3_transformer_amd.ops.fused import _pack_w_dh, _pack_w_fwd
+    P = _pack_w_fwd(W, mo, miF)
+    ms = timeit(lambda: _C.pairconv_fwd(H, P, Ut, out, mo))
     print(f'fwd    ({di},{do}): {ms:8.3f} ms  {gemm_fl/ms/1e9:7.1f} TF/s (gemm) '
           f'{(gemm_fl+epi_fl)/ms/1e9:7.1f} TF/s (total)')
 
-    Wt = W.t().contiguous()
+    P1 = _pack_w_dh(W, mo, miF)
     dH = torch.zeros(E, K, device=dev)
-    ms = timeit(lambda: _C.pairconv_bwd_dh(gt, Ut, Wt, dH, mo))
+    ms = timeit(lambda: _C.pairconv_bwd_dh(gt, Ut, P1, dH, mo))
     print(f'bwd_dh ({di},{do}): {ms:8.3f} ms  {gemm_fl/ms/1e9:7.1f} TF/s')
 
     Ht = H.t().contiguous()
@@ -66,7 +68,7 @@ def main():
     print(f'bwd_dw ({di},{do}): {ms:8.3f} ms  {gemm_fl/ms/1e9:7.1f} TF/s')
 
     dU = torch.empty(miF, O, E, device=dev)
-    ms = timeit(lambda: _C.pairconv_bwd_du(H, W, bias, gt, dU, mo))
+    ms = timeit(lambda: _C.pairconv_bwd_du(H, P, bias, gt, dU, mo))
     print(f'bwd_du ({di},{do}): {ms:8.3f} ms  {gemm_fl/ms/1e9:7.1f} TF/s')
 
     # library GEMM reference: the raw R = H @ W^T (what eager must do, without

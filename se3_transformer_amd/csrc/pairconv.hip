@@ -47,7 +47,7 @@ __device__ __forceinline__ float bf16_to_f32(unsigned short u) {
 template <int O>
 __global__ void __launch_bounds__(NTHREADS)
 pairconv_fwd_kernel(const __bf16* __restrict__ H,
-                    const __bf16* __restrict__ W,
+                    const __bf16* __restrict__ P,   // packed W: [mo/8][miF/32][wm4][mf4][kit4][lane64][8]
                     const __bf16* __restrict__ Ut,
                     float* __restrict__ out,
                     int E, int mo, int miF) {
@@ -114,15 +114,15 @@ pairconv_fwd_kernel(const __bf16* __restrict__ H,
 #pragma unroll
             for (int ef = 0; ef < 2; ++ef) acc[mf][ef] = f32x4(0.f);
 
+        const __bf16* pbase = P + ((((size_t)blockIdx.y * (miF / 32) + c) * 4 + wm) * 4) * 4 * 64 * 8
+                              + (size_t)lane * 8;
 #pragma unroll
         for (int kit = 0; kit < 4; ++kit) {
             const int k0 = kit * 32 + l4 * 8;
             bf16x8 a[4], b[2];
 #pragma unroll
             for (int mf = 0; mf < 4; ++mf) {
-                int r = wm * 64 + mf * 16 + l15;           // n-row in tile
-                size_t n = (size_t)(mo0 + (r >> 5)) * miF + uc0 + (r & 31);
-                a[mf] = *reinterpret_cast<const bf16x8*>(W + n * KDIM + k0);
+                a[mf] = *reinterpret_cast<const bf16x8*>(pbase + ((size_t)mf * 4 + kit) * 64 * 8);
             }
 #pragma unroll
             for (int ef = 0; ef < 2; ++ef) {
@@ -220,7 +220,7 @@ static void launch_fwd(const torch::Tensor& H, const torch::Tensor& W,
                        reinterpret_cast<const __bf16*>(H.data_ptr()),
                        reinterpret_cast<const __bf16*>(W.data_ptr()),
                        reinterpret_cast<const __bf16*>(Ut.data_ptr()),
-                       out.data_ptr<float>(), E, mo, miF);
+                       out.data_ptr<float>(), E, mo, miF);  // W arg = packed P
 }
 
 void pairconv_fwd(torch::Tensor H, torch::Tensor W, torch::Tensor Ut,
@@ -235,7 +235,7 @@ void pairconv_fwd(torch::Tensor H, torch::Tensor W, torch::Tensor Ut,
     int miF = Ut.size(0);
     int O = Ut.size(1);
     TORCH_CHECK(H.size(1) == KDIM, "radial hidden dim must be 128");
-    TORCH_CHECK(W.size(0) == (int64_t)mo * miF && W.size(1) == KDIM);
+    TORCH_CHECK(W.numel() == (int64_t)mo * miF * KDIM, "expect packed W");
     TORCH_CHECK(Ut.size(2) == E);
     TORCH_CHECK(out.size(0) == E && out.size(1) == mo && out.size(2) == O);
     TORCH_CHECK(miF % UCHUNK == 0, "miF must be a multiple of 32");

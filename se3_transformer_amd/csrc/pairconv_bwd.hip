@@ -34,7 +34,7 @@ template <int O>
 __global__ void __launch_bounds__(NT)
 pairconv_bwd_dh_kernel(const __bf16* __restrict__ Gt,  // (mo, O, E) bf16
                        const __bf16* __restrict__ Ut,  // (miF, O, E) bf16
-                       const __bf16* __restrict__ Wt,  // (128, mo*miF) bf16
+                       const __bf16* __restrict__ P1,  // packed W: [mo/8][miF/32][wk2][kf4][ns8][lane64][8]
                        float* __restrict__ dH,         // (E, 128) f32 (zeroed)
                        int E, int mo, int miF, int nsplit) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -50,7 +50,6 @@ pairconv_bwd_dh_kernel(const __bf16* __restrict__ Gt,  // (mo, O, E) bf16
     const int we = wid >> 1;          // 0..3: e-group of 16
     const int wk = wid & 1;           // 0..1: k-group of 64
     const int e0 = blockIdx.x * 64;
-    const size_t N = (size_t)mo * miF;
 
     f32x4 acc[4];                      // 16 e x 64 k per wave
 #pragma unroll
@@ -96,7 +95,8 @@ pairconv_bwd_dh_kernel(const __bf16* __restrict__ Gt,  // (mo, O, E) bf16
             }
             __syncthreads();
             // MFMA: dH_tile += dR(64e x 256n) @ W(256n x 128k)
-            const size_t nbase = (size_t)(mb * 8) * miF + cb * 32;
+            const __bf16* pbase = P1 + ((((size_t)mb * (miF / 32) + cb) * 2 + wk) * 4) * 8 * 64 * 8
+                                  + (size_t)lane * 8;
 #pragma unroll
             for (int ns = 0; ns < 8; ++ns) {      // 8 n-steps of 32
                 // A: dR rows e = we*16+l15, n = ns*32 + l4*8.. (mo-major inside tile)
@@ -108,13 +108,8 @@ pairconv_bwd_dh_kernel(const __bf16* __restrict__ Gt,  // (mo, O, E) bf16
                 // careful: dR tile n index = m*32+c, global n = (mb*8+m)*miF + cb*32 + c
 #pragma unroll
                 for (int kf = 0; kf < 4; ++kf) {
-                    int k = wk * 64 + kf * 16 + l15;
-                    // B[n][k] = Wt[k][n]: 8 consecutive n... but global n is NOT
-                    // contiguous across m boundaries; ns*32+l4*8 stays within one m
-                    // (32-aligned chunks of 32), so c-run of 8 is contiguous.
-                    int m = ntile >> 5, c = ntile & 31;
-                    const __bf16* src = Wt + (size_t)k * N + nbase + (size_t)m * miF + c;
-                    bf16x8 b = *reinterpret_cast<const bf16x8*>(src);
+                    bf16x8 b = *reinterpret_cast<const bf16x8*>(
+                        pbase + ((size_t)kf * 8 + ns) * 64 * 8);
                     acc[kf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(b, a, acc[kf], 0, 0, 0);
                 }
             }
@@ -256,7 +251,7 @@ pairconv_bwd_dw_kernel(const __bf16* __restrict__ Gt,  // (mo, O, E)
 template <int O>
 __global__ void __launch_bounds__(NT)
 pairconv_bwd_du_kernel(const __bf16* __restrict__ H,   // (E,128)
-                       const __bf16* __restrict__ W,   // (mo*miF,128)
+                       const __bf16* __restrict__ P,   // packed W as forward
                        const float* __restrict__ bias, // (mo*miF,)
                        const __bf16* __restrict__ Gt,  // (mo,O,E)
                        float* __restrict__ dU,         // (miF, O, E) f32
@@ -314,15 +309,15 @@ pairconv_bwd_du_kernel(const __bf16* __restrict__ H,   // (E,128)
         for (int mf = 0; mf < 4; ++mf)
 #pragma unroll
             for (int ef = 0; ef < 2; ++ef) acc[mf][ef] = {0.f, 0.f, 0.f, 0.f};
+        const __bf16* pbase = P + ((((size_t)mb * (miF / 32) + cb) * 4 + wm) * 4) * 4 * 64 * 8
+                              + (size_t)lane * 8;
 #pragma unroll
         for (int kit = 0; kit < 4; ++kit) {
             const int k0 = kit * 32 + l4 * 8;
             bf16x8 a[4], b[2];
 #pragma unroll
             for (int mf = 0; mf < 4; ++mf) {
-                int r = wm * 64 + mf * 16 + l15;
-                size_t n = (size_t)(mb * 8 + (r >> 5)) * miF + uc0 + (r & 31);
-                a[mf] = *reinterpret_cast<const bf16x8*>(W + n * KDIM + k0);
+                a[mf] = *reinterpret_cast<const bf16x8*>(pbase + ((size_t)mf * 4 + kit) * 64 * 8);
             }
 #pragma unroll
             for (int ef = 0; ef < 2; ++ef) {
@@ -388,7 +383,7 @@ void pairconv_bwd_dh(torch::Tensor Gt, torch::Tensor Ut, torch::Tensor Wt,
                      torch::Tensor dH, int64_t mo_) {
     int E = dH.size(0), mo = (int)mo_, miF = Ut.size(0), O = Ut.size(1);
     TORCH_CHECK(Gt.is_contiguous() && Ut.is_contiguous() && Wt.is_contiguous() && dH.is_contiguous());
-    TORCH_CHECK(Wt.size(0) == KDIM && Wt.size(1) == (int64_t)mo * miF);
+    TORCH_CHECK(Wt.numel() == (int64_t)mo * miF * KDIM, "expect packed W (P1)");
     auto stream = at::cuda::getCurrentHIPStream();
     int eblk = (E + 63) / 64;
     int nmo = mo / 8;
@@ -433,6 +428,7 @@ void pairconv_bwd_du(torch::Tensor H, torch::Tensor W, torch::Tensor bias,
     int E = H.size(0), mo = (int)mo_, miF = dU.size(0), O = dU.size(1);
     TORCH_CHECK(H.is_contiguous() && W.is_contiguous() && Gt.is_contiguous() &&
                 bias.is_contiguous() && dU.is_contiguous());
+    TORCH_CHECK(W.numel() == (int64_t)mo * miF * KDIM, "expect packed W (P)");
     TORCH_CHECK(bias.dtype() == torch::kFloat32);
     auto stream = at::cuda::getCurrentHIPStream();
     dim3 grid((E + 63) / 64, miF / 32);

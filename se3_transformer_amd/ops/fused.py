@@ -48,6 +48,21 @@ def fused_shapes_ok(mo: int, miF: int, O: int, mid_dim: int) -> bool:
     return mid_dim == 128 and miF % 32 == 0 and mo % 8 == 0 and O in (1, 3, 5, 7)
 
 
+def _pack_w_fwd(W16, mo, miF):
+    """Rearrange W (mo*miF, 128) bf16 into the forward/du kernels' per-lane
+    fragment order [mo/8][miF/32][wm4][mf4][kit4][lane64][j8] so each wave's
+    MFMA A-fragment load is one contiguous 1 KiB read."""
+    v = W16.view(mo // 8, 4, 2, miF // 32, 2, 16, 4, 4, 8)
+    return v.permute(0, 3, 1, 2, 4, 6, 7, 5, 8).contiguous()
+
+
+def _pack_w_dh(W16, mo, miF):
+    """Fragment order for the dH kernel's B-operand:
+    [mo/8][miF/32][wk2][kf4][ns8][lane64][j8]."""
+    v = W16.view(mo // 8, 8, miF // 32, 4, 8, 2, 4, 16)
+    return v.permute(0, 2, 5, 6, 1, 3, 7, 4).contiguous()
+
+
 class _FusedPairConv(torch.autograd.Function):
     """out[e,mo,o] = sum_{c,h} (H[e,h] W[(mo,c),h] + bias[(mo,c)]) * Ut[c,o,e]
 
@@ -67,7 +82,7 @@ class _FusedPairConv(torch.autograd.Function):
         b16 = bias.detach().to(torch.bfloat16).view(mo, miF)
         out = (b16 @ Ut16.reshape(miF, O * E)).view(mo, O, E) \
             .permute(2, 0, 1).contiguous().float()
-        ext.pairconv_fwd(H16, W16, Ut16, out, mo)
+        ext.pairconv_fwd(H16, _pack_w_fwd(W16, mo, miF), Ut16, out, mo)
         ctx.save_for_backward(H16, W16, Ut16, b16)
         ctx.mo = mo
         ctx.w_dtype = W.dtype
@@ -90,10 +105,10 @@ class _FusedPairConv(torch.autograd.Function):
             dH = dW = db = dUt = None
             g_t = g16.permute(1, 2, 0).contiguous()    # (mo, O, E) bf16
             if need_H:
-                Wt = W16.t().contiguous()              # (128, N)
+                P1 = _pack_w_dh(W16, mo, miF)
                 dH = torch.zeros(E, K, dtype=torch.float32, device=H16.device)
-                ext.pairconv_bwd_dh(g_t, Ut16, Wt, dH, mo)
-                del Wt
+                ext.pairconv_bwd_dh(g_t, Ut16, P1, dH, mo)
+                del P1
             if need_W:
                 Ht = H16.t().contiguous()              # (128, E)
                 dW = torch.empty(mo * miF, K, dtype=torch.float32,
@@ -109,8 +124,8 @@ class _FusedPairConv(torch.autograd.Function):
             if need_u:
                 dUt = torch.empty(miF, O, E, dtype=torch.float32,
                                   device=H16.device)
-                ext.pairconv_bwd_du(H16, W16, b16.float().reshape(-1), g_t,
-                                    dUt, mo)
+                ext.pairconv_bwd_du(H16, _pack_w_fwd(W16, mo, miF),
+                                    b16.float().reshape(-1), g_t, dUt, mo)
             return dH, dW, db, dUt, None
 
         u_eco = Ut16.permute(2, 0, 1)                  # (E, miF, O) view
