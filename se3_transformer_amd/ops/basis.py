@@ -142,6 +142,20 @@ def _qj_transposed(J: int, d_in: int, d_out: int) -> torch.Tensor:
     return basis_transformation_Q_J(J, d_in, d_out).t().contiguous()
 
 
+_qj_dev_cache = {}
+
+
+def _qj_t_on(J: int, d_in: int, d_out: int, device, dtype) -> torch.Tensor:
+    """Device/dtype-resident Q_J^T (cached so steady-state forwards — and
+    hipGraph capture — never issue host-to-device copies)."""
+    key = (J, d_in, d_out, device, dtype)
+    t = _qj_dev_cache.get(key)
+    if t is None:
+        t = _qj_transposed(J, d_in, d_out).to(device=device, dtype=dtype)
+        _qj_dev_cache[key] = t
+    return t
+
+
 def _compute_sh(r_ij: torch.Tensor, max_J: int, differentiable: bool) -> torch.Tensor:
     if differentiable:
         return sh_packed_from_cartesian(max_J, r_ij)
@@ -163,7 +177,7 @@ def get_basis(r_ij: torch.Tensor, max_degree: int, differentiable: bool = False)
     for d_in, d_out in product(range(max_degree + 1), range(max_degree + 1)):
         k_js = []
         for J in range(abs(d_in - d_out), d_in + d_out + 1):
-            q_t = _qj_transposed(J, d_in, d_out).to(device=device, dtype=dtype)
+            q_t = _qj_t_on(J, d_in, d_out, device, dtype)
             y_j = y_packed[..., sh_offset(J): sh_offset(J + 1)]
             k_js.append(y_j @ q_t)  # [..., (2do+1)(2di+1)]
         k = torch.stack(k_js, dim=-1)  # [..., (2do+1)(2di+1), F]
@@ -189,7 +203,7 @@ def get_basis_packed(r_ij: torch.Tensor, max_degree: int, differentiable: bool =
     for d_in, d_out in product(range(max_degree + 1), range(max_degree + 1)):
         k_js = []
         for J in range(abs(d_in - d_out), d_in + d_out + 1):
-            q_t = _qj_transposed(J, d_in, d_out).to(device=device, dtype=dtype)
+            q_t = _qj_t_on(J, d_in, d_out, device, dtype)
             y_j = y_packed[..., sh_offset(J): sh_offset(J + 1)]
             k_js.append(y_j @ q_t)
         k = torch.stack(k_js, dim=-1)
