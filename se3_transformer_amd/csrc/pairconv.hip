@@ -147,8 +147,7 @@ pairconv_fwd_kernel(const __bf16* __restrict__ H,
                 for (int o = 0; o < O; ++o) s[ef][mi_][o] = 0.f;
 
 #pragma unroll
-        for (int mf = 0; mf < 4; ++mf) {
-            const int moi = mf >> 1;  // relative mo within the wave's pair
+        for (int mf = 0; mf < 2; ++mf) {   // mf and mf+2 share urow (rows r, r+32)
 #pragma unroll
             for (int reg = 0; reg < 4; ++reg) {
                 const int r = wm * 64 + mf * 16 + l4 * 4 + reg;
@@ -156,12 +155,14 @@ pairconv_fwd_kernel(const __bf16* __restrict__ H,
 #pragma unroll
                 for (int ef = 0; ef < 2; ++ef) {
                     const int e = we * 32 + ef * 16 + l15;
-                    const float rv = acc[mf][ef][reg];
+                    const float rv0 = acc[mf][ef][reg];
+                    const float rv1 = acc[mf + 2][ef][reg];
 #pragma unroll
                     for (int o = 0; o < O; ++o) {
                         float uv = bf16_to_f32(
                             reinterpret_cast<const unsigned short*>(u_lds)[(urow * O + o) * BLK_E + e]);
-                        s[ef][moi][o] = fmaf(rv, uv, s[ef][moi][o]);
+                        s[ef][0][o] = fmaf(rv0, uv, s[ef][0][o]);
+                        s[ef][1][o] = fmaf(rv1, uv, s[ef][1][o]);
                     }
                 }
             }

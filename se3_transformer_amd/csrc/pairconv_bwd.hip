@@ -80,18 +80,25 @@ pairconv_bwd_dh_kernel(const __bf16* __restrict__ Gt,  // (mo, O, E) bf16
                 *reinterpret_cast<bf16x8*>(u_lds + (size_t)ro * 64 + eu) = v;
             }
             __syncthreads();
-            // cooperative dR tile: [64e][256n], each thread 32 elements
-            for (int i = tid; i < 64 * 256; i += NT) {
-                int n = i & 255, e = i >> 8;
-                int m = n >> 5, c = n & 31;
-                float s = 0.f;
+            // cooperative dR tile: [64e][256n]; thread owns (e, run of 8 n):
+            // lanes sweep e so g/u LDS reads are conflict-free, and each run is
+            // one swizzled ds_write_b128.
+            for (int i = tid; i < (64 * 256) / 8; i += NT) {
+                int e = i & 63, n8 = i >> 6;
+                int m = n8 >> 2;                      // (n8*8)>>5
+                __bf16 vals[8];
 #pragma unroll
-                for (int o = 0; o < O; ++o)
-                    s = fmaf(b2f(g_lds[(m * O + o) * 64 + e]),
-                             b2f(u_lds[(c * O + o) * 64 + e]), s);
-                // swizzled: 16B slot (n>>3) xor'd with (e&15) within the row
-                reinterpret_cast<__bf16*>(reinterpret_cast<char*>(dr_lds)
-                    + e * 512 + ((((n >> 3) ^ (e & 15)) << 4)))[n & 7] = (__bf16)s;
+                for (int j = 0; j < 8; ++j) {
+                    int c = (n8 * 8 + j) & 31;
+                    float s = 0.f;
+#pragma unroll
+                    for (int o = 0; o < O; ++o)
+                        s = fmaf(b2f(g_lds[(m * O + o) * 64 + e]),
+                                 b2f(u_lds[(c * O + o) * 64 + e]), s);
+                    vals[j] = (__bf16)s;
+                }
+                *reinterpret_cast<bf16x8*>(reinterpret_cast<char*>(dr_lds)
+                    + e * 512 + (((n8 ^ (e & 15)) << 4))) = *reinterpret_cast<bf16x8*>(vals);
             }
             __syncthreads();
             // MFMA: dH_tile += dR(64e x 256n) @ W(256n x 128k)
