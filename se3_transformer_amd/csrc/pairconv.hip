@@ -45,7 +45,7 @@ __device__ __forceinline__ float bf16_to_f32(unsigned short u) {
 }
 
 template <int O>
-__global__ void __launch_bounds__(NTHREADS)
+__global__ void __launch_bounds__(NTHREADS, 4)   // cap VGPR<=128: 2 blocks/CU
 pairconv_fwd_kernel(const __bf16* __restrict__ H,
                     const __bf16* __restrict__ P,   // packed W: [mo/8][miF/32][wm4][mf4][kit4][lane64][8]
                     const __bf16* __restrict__ Ut,
