@@ -151,10 +151,10 @@ pairconv_bwd_dw_kernel(const __bf16* __restrict__ Gt,  // (mo, O, E)
                        float* __restrict__ dW,         // (mo*miF, 128) f32
                        int E, int mo, int miF) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
-    __bf16* dr_lds = reinterpret_cast<__bf16*>(smem);                  // [128n][64e] 16 KiB
-    __bf16* h_lds = reinterpret_cast<__bf16*>(smem + 16384);           // [128k][64e] 16 KiB
-    __bf16* u_lds = reinterpret_cast<__bf16*>(smem + 32768);           // [32][O][64]
-    __bf16* g_lds = reinterpret_cast<__bf16*>(smem + 32768 + 32 * O * 64 * 2); // [4][O][64]
+    __bf16* dr_lds = reinterpret_cast<__bf16*>(smem);                  // [128n][32e] 8 KiB
+    __bf16* h_lds = reinterpret_cast<__bf16*>(smem + 8192);            // [128k][32e] 8 KiB
+    __bf16* u_lds = reinterpret_cast<__bf16*>(smem + 16384);           // [32][O][32]
+    __bf16* g_lds = reinterpret_cast<__bf16*>(smem + 16384 + 32 * O * 32 * 2); // [4][O][32]
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -173,67 +173,60 @@ pairconv_bwd_dw_kernel(const __bf16* __restrict__ Gt,  // (mo, O, E)
 #pragma unroll
         for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-    const int nec = (E + 63) / 64;
+    const int nec = (E + 31) / 32;
     for (int ec = 0; ec < nec; ++ec) {
-        const int e0 = ec * 64;
+        const int e0 = ec * 32;
         __syncthreads();
-        // stage u chunk [32c][O][64e] and g tile [4m][O][64e] (16B units)
-        for (int i = tid; i < (32 * O * 64) / 8; i += NT) {
-            int ro = i >> 3, eu = (i & 7) * 8;
+        // stage u chunk [32c][O][32e] and g tile [4m][O][32e] (16B units)
+        for (int i = tid; i < (32 * O * 32) / 8; i += NT) {
+            int ro = i >> 2, eu = (i & 3) * 8;
             const __bf16* src = Ut + ((size_t)(cb * 32 + ro / O) * O + (ro % O)) * E + e0 + eu;
             bf16x8 v;
             if (e0 + eu + 8 <= E) v = *reinterpret_cast<const bf16x8*>(src);
             else { for (int j = 0; j < 8; ++j) v[j] = (e0 + eu + j < E) ? src[j] : (__bf16)0.f; }
-            *reinterpret_cast<bf16x8*>(u_lds + (size_t)ro * 64 + eu) = v;
+            *reinterpret_cast<bf16x8*>(u_lds + (size_t)ro * 32 + eu) = v;
         }
-        for (int i = tid; i < (4 * O * 64) / 8; i += NT) {
-            int ro = i >> 3, eu = (i & 7) * 8;
+        for (int i = tid; i < (4 * O * 32) / 8; i += NT) {
+            int ro = i >> 2, eu = (i & 3) * 8;
             const __bf16* src = Gt + ((size_t)(mb * 4 + ro / O) * O + (ro % O)) * E + e0 + eu;
             bf16x8 v;
             if (e0 + eu + 8 <= E) v = *reinterpret_cast<const bf16x8*>(src);
             else { for (int j = 0; j < 8; ++j) v[j] = (e0 + eu + j < E) ? src[j] : (__bf16)0.f; }
-            *reinterpret_cast<bf16x8*>(g_lds + (size_t)ro * 64 + eu) = v;
+            *reinterpret_cast<bf16x8*>(g_lds + (size_t)ro * 32 + eu) = v;
         }
-        // stage H^T chunk [128k][64e], 16B slots XOR-swizzled with (k&7)
-        for (int i = tid; i < (128 * 64) / 8; i += NT) {
-            int k = i >> 3, sl = i & 7;
-            int eu = sl * 8;
+        // stage H^T chunk [128k][32e]
+        for (int i = tid; i < (128 * 32) / 8; i += NT) {
+            int k = i >> 2, eu = (i & 3) * 8;
             const __bf16* src = Ht + (size_t)k * E + e0 + eu;
             bf16x8 v;
             if (e0 + eu + 8 <= E) v = *reinterpret_cast<const bf16x8*>(src);
             else { for (int j = 0; j < 8; ++j) v[j] = (e0 + eu + j < E) ? src[j] : (__bf16)0.f; }
-            *reinterpret_cast<bf16x8*>(reinterpret_cast<char*>(h_lds)
-                + k * 128 + ((sl ^ (k & 7)) << 4)) = v;
+            *reinterpret_cast<bf16x8*>(h_lds + (size_t)k * 32 + eu) = v;
         }
         __syncthreads();
-        // cooperative dR^T tile [128n][64e]
-        for (int i = tid; i < 128 * 64; i += NT) {
-            int e = i & 63, n = i >> 6;
+        // cooperative dR^T tile [128n][32e] + db accumulation
+        for (int i = tid; i < 128 * 32; i += NT) {
+            int e = i & 31, n = i >> 5;
             int m = n >> 5, c = n & 31;
             float s = 0.f;
 #pragma unroll
             for (int o = 0; o < O; ++o)
-                s = fmaf(b2f(g_lds[(m * O + o) * 64 + e]),
-                         b2f(u_lds[(c * O + o) * 64 + e]), s);
-            reinterpret_cast<__bf16*>(reinterpret_cast<char*>(dr_lds)
-                + n * 128 + ((((e >> 3) ^ (n & 7)) << 4)))[e & 7] = (__bf16)s;
+                s = fmaf(b2f(g_lds[(m * O + o) * 32 + e]),
+                         b2f(u_lds[(c * O + o) * 32 + e]), s);
+            dr_lds[(size_t)n * 32 + e] = (__bf16)s;
         }
         __syncthreads();
-        // MFMA: dW_tile += dR^T(128n x 64e) @ H(64e x 128k)
+        // MFMA: dW_tile += dR^T(128n x 32e) @ H(32e x 128k)
 #pragma unroll
-        for (int es = 0; es < 2; ++es) {
+        for (int es = 0; es < 1; ++es) {
 #pragma unroll
             for (int nf = 0; nf < 2; ++nf) {
-                int n = wn * 32 + nf * 16 + l15;
                 bf16x8 a = *reinterpret_cast<const bf16x8*>(
-                    reinterpret_cast<char*>(dr_lds) + n * 128
-                    + (((es * 4 + l4) ^ (n & 7)) << 4));
+                    dr_lds + (size_t)(wn * 32 + nf * 16 + l15) * 32 + l4 * 8);
 #pragma unroll
                 for (int kf = 0; kf < 4; ++kf) {
-                    int k = wk * 64 + kf * 16 + l15;
                     bf16x8 b = *reinterpret_cast<const bf16x8*>(
-                        reinterpret_cast<char*>(h_lds) + k * 128
-                        + (((es * 4 + l4) ^ (k & 7)) << 4));
+                        h_lds + (size_t)(wk * 64 + kf * 16 + l15) * 32 + l4 * 8);
                     acc[nf][kf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nf][kf], 0, 0, 0);
                 }
             }
@@ -426,7 +419,7 @@ void pairconv_bwd_dw(torch::Tensor Gt, torch::Tensor Ut, torch::Tensor Ht,
     auto stream = at::cuda::getCurrentHIPStream();
     dim3 grid((mo / 4) * (miF / 32));
     DISPATCH_O(O, {
-        size_t lds = 32768 + (size_t)32 * kO * 64 * 2 + (size_t)4 * kO * 64 * 2;
+        size_t lds = 16384 + (size_t)32 * kO * 32 * 2 + (size_t)4 * kO * 32 * 2;
         hipLaunchKernelGGL(HIP_KERNEL_NAME(pairconv_bwd_dw_kernel<kO>), grid, dim3(NT), lds, stream,
                            reinterpret_cast<const __bf16*>(Gt.data_ptr()),
                            reinterpret_cast<const __bf16*>(Ut.data_ptr()),
