@@ -58,26 +58,26 @@ pairconv_bwd_dh_kernel(const __bf16* __restrict__ Gt,  // (mo, O, E) bf16
     const int nmo = mo / 8, nuc = miF / 32;
     const int mb_lo = (nmo / nsplit) * blockIdx.y;
     const int mb_hi = mb_lo + nmo / nsplit;
-    for (int mb = mb_lo; mb < mb_hi; ++mb) {
-        // stage g tile [8][O][64] from Gt (e-contiguous rows)
-        for (int i = tid; i < (8 * O * 64) / 8; i += NT) {
-            int ro = i >> 3, eu = (i & 7) * 8;   // ro = m*O+o
-            const __bf16* src = Gt + ((size_t)(mb * 8 + ro / O) * O + (ro % O)) * E + e0 + eu;
+    for (int cb = 0; cb < nuc; ++cb) {
+        // stage u chunk [32][O][64] once per urow-chunk
+        __syncthreads();
+        for (int i = tid; i < (32 * O * 64) / 8; i += NT) {
+            int ro = i >> 3, eu = (i & 7) * 8;
+            const __bf16* src = Ut + ((size_t)(cb * 32 + ro / O) * O + (ro % O)) * E + e0 + eu;
             bf16x8 v;
             if (e0 + eu + 8 <= E) v = *reinterpret_cast<const bf16x8*>(src);
             else { for (int j = 0; j < 8; ++j) v[j] = (e0 + eu + j < E) ? src[j] : (__bf16)0.f; }
-            *reinterpret_cast<bf16x8*>(g_lds + (size_t)ro * 64 + eu) = v;
+            *reinterpret_cast<bf16x8*>(u_lds + (size_t)ro * 64 + eu) = v;
         }
-        for (int cb = 0; cb < nuc; ++cb) {
-            // stage u chunk [32][O][64]
-            __syncthreads();
-            for (int i = tid; i < (32 * O * 64) / 8; i += NT) {
-                int ro = i >> 3, eu = (i & 7) * 8;
-                const __bf16* src = Ut + ((size_t)(cb * 32 + ro / O) * O + (ro % O)) * E + e0 + eu;
+        for (int mb = mb_lo; mb < mb_hi; ++mb) {
+            // stage g tile [8][O][64] from Gt (e-contiguous rows)
+            for (int i = tid; i < (8 * O * 64) / 8; i += NT) {
+                int ro = i >> 3, eu = (i & 7) * 8;   // ro = m*O+o
+                const __bf16* src = Gt + ((size_t)(mb * 8 + ro / O) * O + (ro % O)) * E + e0 + eu;
                 bf16x8 v;
                 if (e0 + eu + 8 <= E) v = *reinterpret_cast<const bf16x8*>(src);
                 else { for (int j = 0; j < 8; ++j) v[j] = (e0 + eu + j < E) ? src[j] : (__bf16)0.f; }
-                *reinterpret_cast<bf16x8*>(u_lds + (size_t)ro * 64 + eu) = v;
+                *reinterpret_cast<bf16x8*>(g_lds + (size_t)ro * 64 + eu) = v;
             }
             __syncthreads();
             // cooperative dR tile: [64e][256n]; thread owns (e, run of 8 n):
