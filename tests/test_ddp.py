@@ -77,3 +77,41 @@ def test_ddp_grad_parity_gloo():
         assert torch.allclose(a, b, atol=1e-5), (name, (a - b).abs().max())
     for name in set(ddp_grads) - set(ref_grads):
         assert ddp_grads[name].abs().max() == 0, name
+
+
+def _worker_bf16(rank, world, tmpdir):
+    import os
+    os.environ.update(MASTER_ADDR='127.0.0.1', MASTER_PORT='29617',
+                      RANK=str(rank), WORLD_SIZE=str(world), LOCAL_RANK=str(rank))
+    import torch
+    import torch.distributed as dist
+    dist.init_process_group('gloo', rank=rank, world_size=world)
+    from se3_transformer_amd.parallel import DistributedDataParallelSE3
+    torch.manual_seed(0)
+    m = torch.nn.Sequential(torch.nn.Linear(16, 16), torch.nn.Linear(16, 4))
+    ddp = DistributedDataParallelSE3(m, grad_compression='bf16')
+    torch.manual_seed(100 + rank)
+    x = torch.randn(8, 16)
+    ddp.zero_grad_buffers()
+    ddp(x).pow(2).mean().backward()
+    ddp.finalize()
+    g = torch.cat([p.grad.flatten() for p in m.parameters()])
+    torch.save(g, f'{tmpdir}/bf16_g{rank}.pt')
+    dist.destroy_process_group()
+
+
+def test_ddp_bf16_compression_gloo(tmp_path):
+    import torch.multiprocessing as mp
+    world = 2
+    ctx = mp.get_context('spawn')
+    ps = [ctx.Process(target=_worker_bf16, args=(r, world, str(tmp_path)))
+          for r in range(world)]
+    for p in ps:
+        p.start()
+    for p in ps:
+        p.join(240)
+        assert p.exitcode == 0
+    import torch
+    g0 = torch.load(tmp_path / 'bf16_g0.pt')
+    g1 = torch.load(tmp_path / 'bf16_g1.pt')
+    assert torch.equal(g0, g1)  # both ranks hold the same reduced grads
