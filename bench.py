@@ -54,8 +54,10 @@ def main():
     p.add_argument('--num-neighbors', type=int, default=8)
     p.add_argument('--valid-radius', type=float, default=10.)
     p.add_argument('--dtype', type=str, default='bf16', choices=['bf16', 'fp32'])
-    p.add_argument('--graph', action='store_true',
-                   help='capture the train step in a hipGraph and replay')
+    p.add_argument('--graph', dest='graph', action='store_true', default=None,
+                   help='capture the train step in a hipGraph and replay '
+                        '(default: on for single-GPU CUDA runs)')
+    p.add_argument('--no-graph', dest='graph', action='store_false')
     p.add_argument('--device', type=str, default=None)
     args = p.parse_args()
 
@@ -105,7 +107,8 @@ def main():
 
     # hipGraph capture: the step is shape-static, so capture once and replay —
     # removes the Python/launch-gap overhead between the ~10k kernels per step.
-    graph_mode = args.graph and use_cuda and world == 1
+    graph_mode = (args.graph if args.graph is not None else world == 1) \
+        and use_cuda and world == 1
     if graph_mode:
         for _ in range(2):          # allocator warmup before capture
             train_step()
