@@ -24,6 +24,13 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 __device__ __forceinline__ float b2f(__bf16 x) { return (float)x; }
 
+typedef __attribute__((ext_vector_type(2))) float f32x2;
+typedef __attribute__((ext_vector_type(2))) __bf16 bf16x2;
+
+__device__ __forceinline__ f32x2 b2f2(bf16x2 v) {
+    return f32x2{(float)v[0], (float)v[1]};
+}
+
 // ---------------------------------------------------------------------------
 // B1: dH[e,k] = sum_n dR[e,n] W[n,k]
 // block: 64 e x 128 k, loops (mo-block 8) x (urow-chunk 32) over all n.
@@ -80,25 +87,34 @@ pairconv_bwd_dh_kernel(const __bf16* __restrict__ Gt,  // (mo, O, E) bf16
                 *reinterpret_cast<bf16x8*>(g_lds + (size_t)ro * 64 + eu) = v;
             }
             __syncthreads();
-            // cooperative dR tile: [64e][256n]; thread owns (e, run of 8 n):
-            // lanes sweep e so g/u LDS reads are conflict-free, and each run is
-            // one swizzled ds_write_b128.
-            for (int i = tid; i < (64 * 256) / 8; i += NT) {
-                int e = i & 63, n8 = i >> 6;
+            // cooperative dR tile: [64e][256n]; thread owns (e-pair, run of 8 n):
+            // b32 LDS reads over e-pairs + packed fma, two swizzled b128 writes.
+            for (int i = tid; i < (64 * 256) / 16; i += NT) {
+                int e = (i & 31) * 2, n8 = i >> 5;
                 int m = n8 >> 2;                      // (n8*8)>>5
-                __bf16 vals[8];
+                f32x2 gv[O];
+#pragma unroll
+                for (int o = 0; o < O; ++o)
+                    gv[o] = b2f2(*reinterpret_cast<const bf16x2*>(g_lds + (m * O + o) * 64 + e));
+                __bf16 v0[8], v1[8];
 #pragma unroll
                 for (int j = 0; j < 8; ++j) {
                     int c = (n8 * 8 + j) & 31;
-                    float s = 0.f;
+                    f32x2 acc2 = {0.f, 0.f};
 #pragma unroll
-                    for (int o = 0; o < O; ++o)
-                        s = fmaf(b2f(g_lds[(m * O + o) * 64 + e]),
-                                 b2f(u_lds[(c * O + o) * 64 + e]), s);
-                    vals[j] = (__bf16)s;
+                    for (int o = 0; o < O; ++o) {
+                        f32x2 uv = b2f2(*reinterpret_cast<const bf16x2*>(
+                            u_lds + (c * O + o) * 64 + e));
+                        acc2[0] = fmaf(gv[o][0], uv[0], acc2[0]);
+                        acc2[1] = fmaf(gv[o][1], uv[1], acc2[1]);
+                    }
+                    v0[j] = (__bf16)acc2[0];
+                    v1[j] = (__bf16)acc2[1];
                 }
                 *reinterpret_cast<bf16x8*>(reinterpret_cast<char*>(dr_lds)
-                    + e * 512 + (((n8 ^ (e & 15)) << 4))) = *reinterpret_cast<bf16x8*>(vals);
+                    + e * 512 + (((n8 ^ (e & 15)) << 4))) = *reinterpret_cast<bf16x8*>(v0);
+                *reinterpret_cast<bf16x8*>(reinterpret_cast<char*>(dr_lds)
+                    + (e + 1) * 512 + (((n8 ^ ((e + 1) & 15)) << 4))) = *reinterpret_cast<bf16x8*>(v1);
             }
             __syncthreads();
             // MFMA: dH_tile += dR(64e x 256n) @ W(256n x 128k)
@@ -204,16 +220,20 @@ pairconv_bwd_dw_kernel(const __bf16* __restrict__ Gt,  // (mo, O, E)
             *reinterpret_cast<bf16x8*>(h_lds + (size_t)k * 32 + eu) = v;
         }
         __syncthreads();
-        // cooperative dR^T tile [128n][32e] + db accumulation
-        for (int i = tid; i < 128 * 32; i += NT) {
-            int e = i & 31, n = i >> 5;
+        // cooperative dR^T tile [128n][32e] (e-pairs, packed)
+        for (int i = tid; i < (128 * 32) / 2; i += NT) {
+            int e = (i & 15) * 2, n = i >> 4;
             int m = n >> 5, c = n & 31;
-            float s = 0.f;
+            f32x2 acc2 = {0.f, 0.f};
 #pragma unroll
-            for (int o = 0; o < O; ++o)
-                s = fmaf(b2f(g_lds[(m * O + o) * 32 + e]),
-                         b2f(u_lds[(c * O + o) * 32 + e]), s);
-            dr_lds[(size_t)n * 32 + e] = (__bf16)s;
+            for (int o = 0; o < O; ++o) {
+                f32x2 gv = b2f2(*reinterpret_cast<const bf16x2*>(g_lds + (m * O + o) * 32 + e));
+                f32x2 uv = b2f2(*reinterpret_cast<const bf16x2*>(u_lds + (c * O + o) * 32 + e));
+                acc2[0] = fmaf(gv[0], uv[0], acc2[0]);
+                acc2[1] = fmaf(gv[1], uv[1], acc2[1]);
+            }
+            *reinterpret_cast<bf16x2*>(dr_lds + (size_t)n * 32 + e) =
+                bf16x2{(__bf16)acc2[0], (__bf16)acc2[1]};
         }
         __syncthreads();
         // MFMA: dW_tile += dR^T(128n x 32e) @ H(32e x 128k)
@@ -351,17 +371,25 @@ pairconv_bwd_du_kernel(const __bf16* __restrict__ H,   // (E,128)
                 }
         __syncthreads();
         // contraction: du_acc[c][o][e] += sum_m R[(m,c)][e] * g[m][o][e]
-        // partition (c,e) across all 512 threads: 32*64 = 2048 cells, 4 per thread
-        for (int i = tid; i < 32 * 64; i += NT) {
-            int e = i & 63, c = i >> 6;
+        // (e-pairs: b32 reads, b64 accumulator RMW; cells owned exclusively)
+        for (int i = tid; i < (32 * 64) / 2; i += NT) {
+            int e = (i & 31) * 2, c = i >> 5;
+            f32x2 rv[8];
+#pragma unroll
+            for (int m = 0; m < 8; ++m)
+                rv[m] = b2f2(*reinterpret_cast<const bf16x2*>(
+                    r_lds + (size_t)(m * 32 + c) * 64 + e));
 #pragma unroll
             for (int o = 0; o < O; ++o) {
-                float s = du_acc[(c * O + o) * 64 + e];
+                f32x2 s2 = *reinterpret_cast<const f32x2*>(du_acc + (c * O + o) * 64 + e);
 #pragma unroll
-                for (int m = 0; m < 8; ++m)
-                    s = fmaf(b2f(r_lds[(size_t)(m * 32 + c) * 64 + e]),
-                             b2f(g_lds[(m * O + o) * 64 + e]), s);
-                du_acc[(c * O + o) * 64 + e] = s;
+                for (int m = 0; m < 8; ++m) {
+                    f32x2 gv = b2f2(*reinterpret_cast<const bf16x2*>(
+                        g_lds + (m * O + o) * 64 + e));
+                    s2[0] = fmaf(rv[m][0], gv[0], s2[0]);
+                    s2[1] = fmaf(rv[m][1], gv[1], s2[1]);
+                }
+                *reinterpret_cast<f32x2*>(du_acc + (c * O + o) * 64 + e) = s2;
             }
         }
         __syncthreads();
