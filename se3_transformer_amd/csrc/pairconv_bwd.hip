@@ -371,25 +371,17 @@ pairconv_bwd_du_kernel(const __bf16* __restrict__ H,   // (E,128)
                 }
         __syncthreads();
         // contraction: du_acc[c][o][e] += sum_m R[(m,c)][e] * g[m][o][e]
-        // (e-pairs: b32 reads, b64 accumulator RMW; cells owned exclusively)
-        for (int i = tid; i < (32 * 64) / 2; i += NT) {
-            int e = (i & 31) * 2, c = i >> 5;
-            f32x2 rv[8];
-#pragma unroll
-            for (int m = 0; m < 8; ++m)
-                rv[m] = b2f2(*reinterpret_cast<const bf16x2*>(
-                    r_lds + (size_t)(m * 32 + c) * 64 + e));
+        // partition (c,e) across all 512 threads: 32*64 = 2048 cells, 4 per thread
+        for (int i = tid; i < 32 * 64; i += NT) {
+            int e = i & 63, c = i >> 6;
 #pragma unroll
             for (int o = 0; o < O; ++o) {
-                f32x2 s2 = *reinterpret_cast<const f32x2*>(du_acc + (c * O + o) * 64 + e);
+                float s = du_acc[(c * O + o) * 64 + e];
 #pragma unroll
-                for (int m = 0; m < 8; ++m) {
-                    f32x2 gv = b2f2(*reinterpret_cast<const bf16x2*>(
-                        g_lds + (m * O + o) * 64 + e));
-                    s2[0] = fmaf(rv[m][0], gv[0], s2[0]);
-                    s2[1] = fmaf(rv[m][1], gv[1], s2[1]);
-                }
-                *reinterpret_cast<f32x2*>(du_acc + (c * O + o) * 64 + e) = s2;
+                for (int m = 0; m < 8; ++m)
+                    s = fmaf(b2f(r_lds[(size_t)(m * 32 + c) * 64 + e]),
+                             b2f(g_lds[(m * O + o) * 64 + e]), s);
+                du_acc[(c * O + o) * 64 + e] = s;
             }
         }
         __syncthreads();
