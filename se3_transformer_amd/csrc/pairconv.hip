@@ -252,6 +252,8 @@ void pairconv_fwd(torch::Tensor H, torch::Tensor W, torch::Tensor Ut,
     TORCH_CHECK(err == hipSuccess, "pairconv_fwd launch failed: ", hipGetErrorString(err));
 }
 
+void sh_basis_fwd(torch::Tensor rel, torch::Tensor qcat, torch::Tensor normtab,
+                  torch::Tensor meta, torch::Tensor out, int64_t L);
 void pairconv_bwd_dh(torch::Tensor G, torch::Tensor Ut, torch::Tensor Wt,
                      torch::Tensor dH, int64_t mo_);
 void pairconv_bwd_dw(torch::Tensor G, torch::Tensor Ut, torch::Tensor Ht,
@@ -265,4 +267,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("pairconv_bwd_dh", &pairconv_bwd_dh, "dH backward");
     m.def("pairconv_bwd_dw", &pairconv_bwd_dw, "dW backward");
     m.def("pairconv_bwd_du", &pairconv_bwd_du, "dU backward");
+    m.def("sh_basis_fwd", &sh_basis_fwd,
+          "fused spherical-harmonics + equivariant basis (MI355X)");
 }

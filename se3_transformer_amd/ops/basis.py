@@ -190,13 +190,65 @@ def get_basis(r_ij: torch.Tensor, max_degree: int, differentiable: bool = False)
     return basis
 
 
+_sh_tables_cache = {}
+
+
+def _sh_basis_tables(max_degree: int, device):
+    """Device-resident Q_J^T concat + per-pair meta + SH norm table for the
+    fused sh_basis HIP kernel (cached; hipGraph-capture safe)."""
+    key = (max_degree, device)
+    hit = _sh_tables_cache.get(key)
+    if hit is not None:
+        return hit
+    L = 2 * max_degree
+    qparts, meta, layout = [], [], {}
+    off_out = off_q = 0
+    for d_in, d_out in product(range(max_degree + 1), range(max_degree + 1)):
+        O, I = to_order(d_out), to_order(d_in)
+        F = num_basis_freq(d_in, d_out)
+        meta.append((d_in, d_out, off_out, off_q))
+        layout[(d_in, d_out)] = (off_out, O, I, F)
+        for J in range(abs(d_in - d_out), d_in + d_out + 1):
+            q_t = _qj_transposed(J, d_in, d_out).float()   # (2J+1, O*I)
+            qparts.append(q_t.reshape(-1))
+            off_q += q_t.numel()
+        off_out += O * I * F
+    from .sh import _norm_const
+    normtab = torch.tensor([_norm_const(l, m) for l in range(L + 1)
+                            for m in range(l + 1)], dtype=torch.float32)
+    tables = (torch.cat(qparts).to(device),
+              normtab.to(device),
+              torch.tensor(meta, dtype=torch.int32, device=device),
+              off_out, layout)
+    _sh_tables_cache[key] = tables
+    return tables
+
+
 def get_basis_packed(r_ij: torch.Tensor, max_degree: int, differentiable: bool = False):
     """Compact per-pair basis for the fused conv path.
 
     Returns {(d_in, d_out): tensor [..., 2*d_out+1, 2*d_in+1, F]} with
     F = 2*min(d_in,d_out)+1, contiguous, no broadcast singleton dims.
+
+    On CUDA fp32 non-differentiable inputs this runs the fused HIP
+    spherical-harmonics + basis kernel (csrc/sh_basis.hip) in one launch.
     """
     device, dtype = r_ij.device, r_ij.dtype
+
+    if (r_ij.is_cuda and not differentiable and dtype == torch.float32
+            and os.environ.get('SE3_EAGER_BASIS') != '1'):
+        from . import fused as _fused
+        if _fused.ext_available():
+            qcat, normtab, meta, TOT, layout = _sh_basis_tables(max_degree, device)
+            rel = r_ij.reshape(-1, 3).contiguous()
+            E = rel.shape[0]
+            out = torch.empty(E, TOT, dtype=torch.float32, device=device)
+            _fused._EXT.sh_basis_fwd(rel, qcat, normtab, meta, out, 2 * max_degree)
+            basis = {}
+            for (d_in, d_out), (off, O, I, F) in layout.items():
+                basis[(d_in, d_out)] = out[:, off:off + O * I * F] \
+                    .reshape(*r_ij.shape[:-1], O, I, F)
+            return basis
     y_packed = _compute_sh(r_ij, 2 * max_degree, differentiable)
 
     basis = {}

@@ -18,7 +18,8 @@ setup(
         CUDAExtension(
             name='se3_transformer_amd._C',
             sources=['se3_transformer_amd/csrc/pairconv.hip',
-                     'se3_transformer_amd/csrc/pairconv_bwd.hip'],
+                     'se3_transformer_amd/csrc/pairconv_bwd.hip',
+                     'se3_transformer_amd/csrc/sh_basis.hip'],
             extra_compile_args={
                 'cxx': ['-O3'],
                 'nvcc': ['-O3', '--offload-arch=gfx950'],

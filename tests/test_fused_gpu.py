@@ -152,3 +152,22 @@ def test_model_fused_bf16_close_to_eager_fp32():
         out = model(feats, coors, mask, return_type=0)
     err = _rel_err(out.float(), ref.float())
     assert err < 0.1, f'bf16 fused vs fp32 eager: {err}'
+
+
+@needs_gpu
+@pytest.mark.parametrize('max_degree', [1, 2, 3])
+def test_sh_basis_kernel_vs_eager(max_degree):
+    import os as _os
+    from se3_transformer_amd.ops.basis import get_basis_packed
+    torch.manual_seed(3)
+    rel = torch.randn(2, 100, 8, 3, device='cuda')
+    _os.environ['SE3_EAGER_BASIS'] = '1'
+    try:
+        ref = get_basis_packed(rel, max_degree)
+    finally:
+        del _os.environ['SE3_EAGER_BASIS']
+    fast = get_basis_packed(rel, max_degree)
+    assert set(ref) == set(fast)
+    for k in ref:
+        err = (ref[k] - fast[k]).abs().max().item()
+        assert err < 1e-4, f'{k}: {err}'
