@@ -254,6 +254,9 @@ void pairconv_fwd(torch::Tensor H, torch::Tensor W, torch::Tensor Ut,
 
 void sh_basis_fwd(torch::Tensor rel, torch::Tensor qcat, torch::Tensor normtab,
                   torch::Tensor meta, torch::Tensor out, int64_t L);
+void norm_se3_fwd(torch::Tensor t, torch::Tensor scale, torch::Tensor out, double eps);
+void norm_se3_bwd(torch::Tensor t, torch::Tensor scale, torch::Tensor dout,
+                  torch::Tensor dt, torch::Tensor dscale, double eps);
 void pairconv_bwd_dh(torch::Tensor G, torch::Tensor Ut, torch::Tensor Wt,
                      torch::Tensor dH, int64_t mo_);
 void pairconv_bwd_dw(torch::Tensor G, torch::Tensor Ut, torch::Tensor Ht,
@@ -267,6 +270,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("pairconv_bwd_dh", &pairconv_bwd_dh, "dH backward");
     m.def("pairconv_bwd_dw", &pairconv_bwd_dw, "dW backward");
     m.def("pairconv_bwd_du", &pairconv_bwd_du, "dU backward");
+    m.def("norm_se3_fwd", &norm_se3_fwd, "fused NormSE3 forward");
+    m.def("norm_se3_bwd", &norm_se3_bwd, "fused NormSE3 backward");
     m.def("sh_basis_fwd", &sh_basis_fwd,
           "fused spherical-harmonics + equivariant basis (MI355X)");
 }

@@ -96,8 +96,19 @@ class NormSE3(nn.Module):
             self.transform[str(degree)] = nn.ParameterDict(entries)
 
     def forward(self, features):
+        from ..ops.fused import ext_available, norm_se3
+        use_fused = (not self.gated_scale and isinstance(self.nonlin, nn.GELU)
+                     and getattr(self.nonlin, 'approximate', 'none') == 'none'
+                     and os.environ.get('SE3_EAGER_NORM') != '1')
         output = {}
         for degree, t in features.items():
+            if (use_fused and t.is_cuda and t.shape[-1] in (1, 3, 5, 7)
+                    and t.dtype in (torch.float32, torch.bfloat16)
+                    and ext_available()):
+                output[degree] = norm_se3(t.contiguous(),
+                                          self.transform[degree]['scale'],
+                                          self.eps)
+                continue
             norm = t.norm(dim=-1, keepdim=True).clamp(min=self.eps)
             phase = t / norm
             params = self.transform[degree]

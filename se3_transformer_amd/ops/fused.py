@@ -170,3 +170,32 @@ class _FusedPairConv(torch.autograd.Function):
 
 def fused_pairconv(H, W, bias, Ut, mo):
     return _FusedPairConv.apply(H, W, bias, Ut, mo)
+
+
+class _NormSE3Fn(torch.autograd.Function):
+    """Fused equivariant norm-gate (csrc/norm_se3.hip), scale path."""
+
+    @staticmethod
+    def forward(ctx, t, scale, eps):
+        ext = _load_ext()
+        s32 = scale.detach().reshape(-1).float().contiguous()
+        out = torch.empty_like(t)
+        ext.norm_se3_fwd(t, s32, out, eps)
+        ctx.save_for_backward(t, s32)
+        ctx.eps = eps
+        ctx.scale_shape = scale.shape
+        ctx.scale_dtype = scale.dtype
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        ext = _load_ext()
+        t, s32 = ctx.saved_tensors
+        dt = torch.empty_like(t)
+        dscale = torch.zeros_like(s32)
+        ext.norm_se3_bwd(t, s32, dout.contiguous(), dt, dscale, ctx.eps)
+        return dt, dscale.view(ctx.scale_shape).to(ctx.scale_dtype), None
+
+
+def norm_se3(t, scale, eps):
+    return _NormSE3Fn.apply(t, scale, eps)

@@ -19,7 +19,8 @@ setup(
             name='se3_transformer_amd._C',
             sources=['se3_transformer_amd/csrc/pairconv.hip',
                      'se3_transformer_amd/csrc/pairconv_bwd.hip',
-                     'se3_transformer_amd/csrc/sh_basis.hip'],
+                     'se3_transformer_amd/csrc/sh_basis.hip',
+                     'se3_transformer_amd/csrc/norm_se3.hip'],
             extra_compile_args={
                 'cxx': ['-O3'],
                 'nvcc': ['-O3', '--offload-arch=gfx950'],

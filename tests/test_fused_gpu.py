@@ -171,3 +171,31 @@ def test_sh_basis_kernel_vs_eager(max_degree):
     for k in ref:
         err = (ref[k] - fast[k]).abs().max().item()
         assert err < 1e-4, f'{k}: {err}'
+
+
+@needs_gpu
+@pytest.mark.parametrize('m', [1, 3, 7])
+def test_norm_se3_kernel_vs_eager(m):
+    import os as _os
+    from se3_transformer_amd.models.core import NormSE3
+    from se3_transformer_amd.models.fiber import Fiber
+    torch.manual_seed(4)
+    degree = (m - 1) // 2
+    norm = NormSE3(Fiber([(degree, 24)])).to('cuda')
+    t0 = torch.randn(3, 50, 24, m, device='cuda', requires_grad=True)
+    t1 = t0.detach().clone().requires_grad_(True)
+
+    _os.environ['SE3_EAGER_NORM'] = '1'
+    try:
+        ref = norm({str(degree): t0})[str(degree)]
+    finally:
+        del _os.environ['SE3_EAGER_NORM']
+    ref.pow(2).mean().backward()
+    out = norm({str(degree): t1})[str(degree)]
+    assert _rel_err(out, ref) < 1e-5
+    gs_ref = norm.transform[str(degree)]['scale'].grad.clone()
+    norm.zero_grad()
+    out.pow(2).mean().backward()
+    gs = norm.transform[str(degree)]['scale'].grad
+    assert _rel_err(t1.grad, t0.grad) < 1e-4
+    assert _rel_err(gs, gs_ref) < 1e-4
