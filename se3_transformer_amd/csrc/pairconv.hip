@@ -254,6 +254,9 @@ void pairconv_fwd(torch::Tensor H, torch::Tensor W, torch::Tensor Ut,
 
 void sh_basis_fwd(torch::Tensor rel, torch::Tensor qcat, torch::Tensor normtab,
                   torch::Tensor meta, torch::Tensor out, int64_t L);
+void attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+              torch::Tensor mask, torch::Tensor out,
+              int64_t n, int64_t heads, double scale);
 void norm_se3_fwd(torch::Tensor t, torch::Tensor scale, torch::Tensor out, double eps);
 void norm_se3_bwd(torch::Tensor t, torch::Tensor scale, torch::Tensor dout,
                   torch::Tensor dt, torch::Tensor dscale, double eps);
@@ -270,6 +273,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("pairconv_bwd_dh", &pairconv_bwd_dh, "dH backward");
     m.def("pairconv_bwd_dw", &pairconv_bwd_dw, "dW backward");
     m.def("pairconv_bwd_du", &pairconv_bwd_du, "dU backward");
+    m.def("attn_fwd", &attn_fwd, "fused neighbor attention forward");
     m.def("norm_se3_fwd", &norm_se3_fwd, "fused NormSE3 forward");
     m.def("norm_se3_bwd", &norm_se3_bwd, "fused NormSE3 backward");
     m.def("sh_basis_fwd", &sh_basis_fwd,

@@ -7,6 +7,8 @@ neighbor-attention kernel on GPU; the einsum path below is the oracle.
 """
 from __future__ import annotations
 
+import os
+
 import torch
 import torch.nn.functional as F
 from torch import nn
@@ -146,15 +148,34 @@ class AttentionSE3(nn.Module):
                 k = torch.cat((global_k, k), dim=3)
                 v = torch.cat((global_v, v), dim=3)
 
-            sim = torch.einsum('bhidm,bhijdm->bhij', q, k) * self.scale
+            bq, hq, nq, dq, mq = q.shape
+            J = k.shape[3]
+            from ..ops import fused as _fused
+            if (q.is_cuda and J <= 64 and dq * mq <= 448
+                    and q.dtype in (torch.float32, torch.bfloat16)
+                    and os.environ.get('SE3_EAGER_ATTN') != '1'
+                    and _fused.ext_available()):
+                mask_u8 = None
+                if neighbor_mask is not None:
+                    pad = J - neighbor_mask.shape[-1]
+                    mask_u8 = F.pad(neighbor_mask, (pad, 0), value=True) \
+                        .squeeze(1).to(torch.uint8).contiguous()
+                out = _fused.fused_attention(
+                    q.reshape(bq * hq * nq, dq * mq).contiguous(),
+                    k.to(q.dtype).reshape(bq * hq * nq, J, dq * mq).contiguous(),
+                    v.to(q.dtype).reshape(bq * hq * nq, J, dq * mq).contiguous(),
+                    mask_u8, nq, hq, self.scale)
+                out = out.view(bq, hq, nq, dq, mq).to(q.dtype)
+            else:
+                sim = torch.einsum('bhidm,bhijdm->bhij', q, k) * self.scale
 
-            if neighbor_mask is not None:
-                num_left_pad = sim.shape[-1] - neighbor_mask.shape[-1]
-                padded_mask = F.pad(neighbor_mask, (num_left_pad, 0), value=True)
-                sim = sim.masked_fill(~padded_mask, -torch.finfo(sim.dtype).max)
+                if neighbor_mask is not None:
+                    num_left_pad = sim.shape[-1] - neighbor_mask.shape[-1]
+                    padded_mask = F.pad(neighbor_mask, (num_left_pad, 0), value=True)
+                    sim = sim.masked_fill(~padded_mask, -torch.finfo(sim.dtype).max)
 
-            attn = sim.softmax(dim=-1)
-            out = torch.einsum('bhij,bhijdm->bhidm', attn, v)
+                attn = sim.softmax(dim=-1)
+                out = torch.einsum('bhij,bhijdm->bhidm', attn, v)
             # b h n d m -> b n (h d) m
             outputs[degree] = out.permute(0, 2, 1, 3, 4).reshape(b, n, -1, m)
 

@@ -199,3 +199,46 @@ class _NormSE3Fn(torch.autograd.Function):
 
 def norm_se3(t, scale, eps):
     return _NormSE3Fn.apply(t, scale, eps)
+
+
+class _AttnFn(torch.autograd.Function):
+    """Fused neighbor attention (csrc/attn.hip): rows (b,h,i), keys J<=64.
+    Backward is the standard softmax-attention adjoint in library ops."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, mask_u8, n, heads, scale):
+        ext = _load_ext()
+        R, DM = q.shape
+        out = torch.empty(R, DM, dtype=torch.float32, device=q.device)
+        m = mask_u8 if mask_u8 is not None else \
+            torch.empty(0, dtype=torch.uint8, device=q.device)
+        ext.attn_fwd(q, k, v, m, out, n, heads, scale)
+        ctx.save_for_backward(q, k, v,
+                              mask_u8 if mask_u8 is not None else None)
+        ctx.dims = (n, heads, scale)
+        return out
+
+    @staticmethod
+    def backward(ctx, g):
+        q, k, v, mask_u8 = ctx.saved_tensors
+        n, heads, scale = ctx.dims
+        R, J, DM = k.shape
+        qf, kf, vf, gf = q.float(), k.float(), v.float(), g.contiguous().float()
+        sim = torch.einsum('rd,rjd->rj', qf, kf) * scale
+        if mask_u8 is not None:
+            b = R // (heads * n)
+            mrows = mask_u8.view(b, 1, n, J).expand(b, heads, n, J) \
+                .reshape(R, J).bool()
+            sim = sim.masked_fill(~mrows, -torch.finfo(sim.dtype).max)
+        attn = sim.softmax(dim=-1)
+        dv = torch.einsum('rj,rd->rjd', attn, gf)
+        dattn = torch.einsum('rd,rjd->rj', gf, vf)
+        dsim = attn * (dattn - (attn * dattn).sum(-1, keepdim=True))
+        dq = torch.einsum('rj,rjd->rd', dsim, kf) * scale
+        dk = torch.einsum('rj,rd->rjd', dsim, qf) * scale
+        return (dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype),
+                None, None, None, None)
+
+
+def fused_attention(q, k, v, mask_u8, n, heads, scale):
+    return _AttnFn.apply(q, k, v, mask_u8, n, heads, scale)

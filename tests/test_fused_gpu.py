@@ -199,3 +199,40 @@ def test_norm_se3_kernel_vs_eager(m):
     gs = norm.transform[str(degree)]['scale'].grad
     assert _rel_err(t1.grad, t0.grad) < 1e-4
     assert _rel_err(gs, gs_ref) < 1e-4
+
+
+@needs_gpu
+def test_fused_attention_vs_eager():
+    import os as _os
+    from se3_transformer_amd.models.attention import AttentionSE3
+    from se3_transformer_amd.models.fiber import Fiber
+    torch.manual_seed(5)
+    fiber = Fiber([(0, 32), (1, 32)])
+    attn = AttentionSE3(fiber, dim_head=16, heads=2, attend_self=True,
+                        use_null_kv=True).to('cuda')
+    b, n, kn = 2, 40, 6
+    feats = {'0': torch.randn(b, n, 32, 1, device='cuda', requires_grad=True),
+             '1': torch.randn(b, n, 32, 3, device='cuda', requires_grad=True)}
+    feats2 = {k: v.detach().clone().requires_grad_(True) for k, v in feats.items()}
+    nbr_idx = torch.randint(0, n, (b, n, kn), device='cuda')
+    nbr_mask = torch.rand(b, n, kn, device='cuda') > 0.2
+    rel_dist = torch.rand(b, n, kn, device='cuda')
+    from se3_transformer_amd.ops.basis import get_basis_packed
+    rel_pos = torch.randn(b, n, kn, 3, device='cuda')
+    basis = get_basis_packed(rel_pos, 1)
+    edge_info = (nbr_idx, nbr_mask, None)
+
+    _os.environ['SE3_EAGER_ATTN'] = '1'
+    try:
+        ref = attn(feats, edge_info, rel_dist, basis)
+    finally:
+        del _os.environ['SE3_EAGER_ATTN']
+    (ref['0'].pow(2).mean() + ref['1'].pow(2).mean()).backward()
+    gref = {k: v.grad.clone() for k, v in feats.items()}
+
+    out = attn(feats2, edge_info, rel_dist, basis)
+    for d in ref:
+        assert _rel_err(out[d], ref[d]) < 1e-4, d
+    (out['0'].pow(2).mean() + out['1'].pow(2).mean()).backward()
+    for d in gref:
+        assert _rel_err(feats2[d].grad, gref[d]) < 1e-3, d
