@@ -7,6 +7,8 @@ masked_select) graph construction, fused conv path, no `splits` chunking.
 """
 from __future__ import annotations
 
+import os
+
 import torch
 import torch.nn.functional as F
 from torch import nn
@@ -291,65 +293,99 @@ class SE3Transformer(nn.Module):
             values, indices = adj_values.topk(num_sparse_neighbors, dim=-1)
             sparse_neighbor_mask = torch.zeros_like(adj_values).scatter_(-1, indices, values) > 0.5
 
-        # relative geometry (self excluded by indexing, not masked_select)
-        indices = off_diag.unsqueeze(0).expand(b, n, n - 1)
-        rel_pos_full = coors.unsqueeze(2) - coors.unsqueeze(1)  # b i j 3
-        rel_pos = _remove_self(rel_pos_full, off_diag)          # b i j-1 3
+        # on-device kNN kernel path (csrc/knn.hip): replaces the dense
+        # (b,n,n) rel-geometry build + masked topk + gathers for the common
+        # configuration; selection semantics identical to the eager branch.
+        from ..ops import fused as _fusedmod
+        k_eff = int(min(neighbors, n - 1))
+        use_knn = (coors.is_cuda and coors.dtype == torch.float32
+                   and not self.differentiable_coors
+                   and sparse_neighbor_mask is None
+                   and not exists(neighbor_mask)
+                   and neighbors > 0 and 1 <= k_eff <= 16
+                   and self.adj_emb is None
+                   and os.environ.get('SE3_EAGER_KNN') != '1'
+                   and _fusedmod.ext_available())
 
-        if exists(mask):
-            mask_pair = mask.unsqueeze(2) & mask.unsqueeze(1)
-            mask_pair = _remove_self(mask_pair, off_diag)
+        if use_knn:
+            idx = torch.empty(b, n, k_eff, dtype=torch.int64, device=device)
+            dist = torch.empty(b, n, k_eff, device=device)
+            relp = torch.empty(b, n, k_eff, 3, device=device)
+            nm = torch.empty(b, n, k_eff, dtype=torch.uint8, device=device)
+            node_mask_u8 = mask.to(torch.uint8).contiguous() if exists(mask) \
+                else torch.empty(0, dtype=torch.uint8, device=device)
+            _fusedmod._EXT.knn_graph(coors.contiguous(), node_mask_u8, idx,
+                                     dist, relp, nm, k_eff,
+                                     float(valid_radius), bool(self.causal))
+            neighbor_indices = idx
+            neighbor_rel_dist = dist
+            neighbor_rel_pos = relp
+            neighbor_mask = nm.bool()
 
-        if exists(edges):
-            if exists(self.edge_emb):
-                edges = self.edge_emb(edges)
-            edges = _remove_self(edges, off_diag)
+            if exists(edges):
+                if exists(self.edge_emb):
+                    edges = self.edge_emb(edges)
+                edges = batched_index_select(edges, neighbor_indices, dim=2)
+        else:
+            # relative geometry (self excluded by indexing, not masked_select)
+            indices = off_diag.unsqueeze(0).expand(b, n, n - 1)
+            rel_pos_full = coors.unsqueeze(2) - coors.unsqueeze(1)  # b i j 3
+            rel_pos = _remove_self(rel_pos_full, off_diag)          # b i j-1 3
 
-        if exists(self.adj_emb):
-            adj_emb = self.adj_emb(adj_indices)
-            edges = torch.cat((edges, adj_emb), dim=-1) if exists(edges) else adj_emb
+            if exists(mask):
+                mask_pair = mask.unsqueeze(2) & mask.unsqueeze(1)
+                mask_pair = _remove_self(mask_pair, off_diag)
 
-        rel_dist = rel_pos.norm(dim=-1)
+            if exists(edges):
+                if exists(self.edge_emb):
+                    edges = self.edge_emb(edges)
+                edges = _remove_self(edges, off_diag)
 
-        # neighbor selection distances
-        modified_rel_dist = rel_dist
-        max_value = torch.finfo(modified_rel_dist.dtype).max
+            if exists(self.adj_emb):
+                adj_emb = self.adj_emb(adj_indices)
+                edges = torch.cat((edges, adj_emb), dim=-1) if exists(edges) else adj_emb
 
-        if exists(neighbor_mask):
-            neighbor_mask = _remove_self(neighbor_mask, off_diag)
-            max_neighbors = int(neighbor_mask.sum(dim=-1).max().item())
-            if max_neighbors > neighbors:
-                print(f'neighbor_mask shows maximum number of neighbors as {max_neighbors} '
-                      f'but specified number of neighbors is {neighbors}')
-            modified_rel_dist = modified_rel_dist.masked_fill(~neighbor_mask, max_value)
+            rel_dist = rel_pos.norm(dim=-1)
 
-        if exists(sparse_neighbor_mask):
-            modified_rel_dist = modified_rel_dist.masked_fill(sparse_neighbor_mask, 0.)
+            # neighbor selection distances
+            modified_rel_dist = rel_dist
+            max_value = torch.finfo(modified_rel_dist.dtype).max
 
-        if self.causal:
-            causal_mask = torch.ones(n, n - 1, device=device).triu().bool()
-            modified_rel_dist = modified_rel_dist.masked_fill(causal_mask.unsqueeze(0), max_value)
+            if exists(neighbor_mask):
+                neighbor_mask = _remove_self(neighbor_mask, off_diag)
+                max_neighbors = int(neighbor_mask.sum(dim=-1).max().item())
+                if max_neighbors > neighbors:
+                    print(f'neighbor_mask shows maximum number of neighbors as {max_neighbors} '
+                          f'but specified number of neighbors is {neighbors}')
+                modified_rel_dist = modified_rel_dist.masked_fill(~neighbor_mask, max_value)
 
-        if neighbors == 0:
-            valid_radius = 0
+            if exists(sparse_neighbor_mask):
+                modified_rel_dist = modified_rel_dist.masked_fill(sparse_neighbor_mask, 0.)
 
-        neighbors = int(min(neighbors, n - 1))
-        total_neighbors = int(neighbors + num_sparse_neighbors)
-        assert total_neighbors > 0, 'must be fetching at least 1 neighbor'
-        total_neighbors = int(min(total_neighbors, n - 1))
+            if self.causal:
+                causal_mask = torch.ones(n, n - 1, device=device).triu().bool()
+                modified_rel_dist = modified_rel_dist.masked_fill(causal_mask.unsqueeze(0), max_value)
 
-        dist_values, nearest_indices = modified_rel_dist.topk(total_neighbors, dim=-1, largest=False)
-        neighbor_mask = dist_values <= valid_radius
+            if neighbors == 0:
+                valid_radius = 0
 
-        neighbor_rel_dist = batched_index_select(rel_dist, nearest_indices, dim=2)
-        neighbor_rel_pos = batched_index_select(rel_pos, nearest_indices, dim=2)
-        neighbor_indices = batched_index_select(indices, nearest_indices, dim=2)
+            neighbors = int(min(neighbors, n - 1))
+            total_neighbors = int(neighbors + num_sparse_neighbors)
+            assert total_neighbors > 0, 'must be fetching at least 1 neighbor'
+            total_neighbors = int(min(total_neighbors, n - 1))
 
-        if exists(mask):
-            neighbor_mask = neighbor_mask & batched_index_select(mask_pair, nearest_indices, dim=2)
+            dist_values, nearest_indices = modified_rel_dist.topk(total_neighbors, dim=-1, largest=False)
+            neighbor_mask = dist_values <= valid_radius
 
-        if exists(edges):
-            edges = batched_index_select(edges, nearest_indices, dim=2)
+            neighbor_rel_dist = batched_index_select(rel_dist, nearest_indices, dim=2)
+            neighbor_rel_pos = batched_index_select(rel_pos, nearest_indices, dim=2)
+            neighbor_indices = batched_index_select(indices, nearest_indices, dim=2)
+
+            if exists(mask):
+                neighbor_mask = neighbor_mask & batched_index_select(mask_pair, nearest_indices, dim=2)
+
+            if exists(edges):
+                edges = batched_index_select(edges, nearest_indices, dim=2)
 
         # rotary embeddings
         rotary_pos_emb = None
