@@ -50,7 +50,7 @@ pairconv_fwd_kernel(const __bf16* __restrict__ H,
                     const __bf16* __restrict__ P,   // packed W: [mo/8][miF/32][wm4][mf4][kit4][lane64][8]
                     const __bf16* __restrict__ Ut,
                     float* __restrict__ out,
-                    int E, int mo, int miF) {
+                    int E, int mo, int miF, int nmemb, int coh) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
     // carve: H tile | u chunk | partial accumulator
     __bf16* h_lds = reinterpret_cast<__bf16*>(smem);                       // [64][128] swizzled, 16 KiB
@@ -65,8 +65,19 @@ pairconv_fwd_kernel(const __bf16* __restrict__ H,
     const int l15 = lane & 15;
     const int l4 = lane >> 4;          // 0..3
 
-    const int e0 = blockIdx.x * BLK_E;
-    const int mo0 = blockIdx.y * BLK_MO;
+    // cohort mapping: all concurrently-resident blocks on one XCD share the
+    // same mo-block (=> the same packed-W slice stays in that XCD's L2).
+    int eb, mb;
+    if (coh) {
+        int x = blockIdx.x & 7, r = blockIdx.x >> 3;
+        eb = r % nmemb;
+        mb = x + 8 * (r / nmemb);
+    } else {
+        eb = blockIdx.x % nmemb;
+        mb = blockIdx.x / nmemb;
+    }
+    const int e0 = eb * BLK_E;
+    const int mo0 = mb * BLK_MO;
 
     // ---- stage H tile (64 x 128 bf16), XOR-swizzled 16B slots within each row
     {
@@ -114,7 +125,7 @@ pairconv_fwd_kernel(const __bf16* __restrict__ H,
 #pragma unroll
             for (int ef = 0; ef < 2; ++ef) acc[mf][ef] = f32x4(0.f);
 
-        const __bf16* pbase = P + ((((size_t)blockIdx.y * (miF / 32) + c) * 4 + wm) * 4) * 4 * 64 * 8
+        const __bf16* pbase = P + ((((size_t)mb * (miF / 32) + c) * 4 + wm) * 4) * 4 * 64 * 8
                               + (size_t)lane * 8;
 #pragma unroll
         for (int kit = 0; kit < 4; ++kit) {
@@ -214,14 +225,17 @@ template <int O>
 static void launch_fwd(const torch::Tensor& H, const torch::Tensor& W,
                        const torch::Tensor& Ut, torch::Tensor& out,
                        int E, int mo, int miF) {
-    dim3 grid((E + BLK_E - 1) / BLK_E, mo / BLK_MO);
+    int nmemb = (E + BLK_E - 1) / BLK_E;
+    int ng = mo / BLK_MO;
+    int coh = (ng % 8 == 0) ? 1 : 0;
+    dim3 grid(nmemb * ng);
     size_t lds = 16384 + (size_t)UCHUNK * O * BLK_E * 2 + (size_t)BLK_E * BLK_MO * O * 4;
     auto stream = at::cuda::getCurrentHIPStream();
     hipLaunchKernelGGL(HIP_KERNEL_NAME(pairconv_fwd_kernel<O>), grid, dim3(NTHREADS), lds, stream,
                        reinterpret_cast<const __bf16*>(H.data_ptr()),
                        reinterpret_cast<const __bf16*>(W.data_ptr()),
                        reinterpret_cast<const __bf16*>(Ut.data_ptr()),
-                       out.data_ptr<float>(), E, mo, miF);  // W arg = packed P
+                       out.data_ptr<float>(), E, mo, miF, nmemb, coh);  // W arg = packed P
 }
 
 void pairconv_fwd(torch::Tensor H, torch::Tensor W, torch::Tensor Ut,

@@ -43,7 +43,7 @@ pairconv_bwd_dh_kernel(const __bf16* __restrict__ Gt,  // (mo, O, E) bf16
                        const __bf16* __restrict__ Ut,  // (miF, O, E) bf16
                        const __bf16* __restrict__ P1,  // packed W: [mo/8][miF/32][wk2][kf4][ns8][lane64][8]
                        float* __restrict__ dH,         // (E, 128) f32 (zeroed)
-                       int E, int mo, int miF, int nsplit) {
+                       int E, int mo, int miF, int nsplit, int nmemb, int coh) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
     __bf16* dr_lds = reinterpret_cast<__bf16*>(smem);                 // [64e][256n] 32 KiB
     __bf16* u_lds = reinterpret_cast<__bf16*>(smem + 32768);          // [32][O][64]
@@ -56,14 +56,23 @@ pairconv_bwd_dh_kernel(const __bf16* __restrict__ Gt,  // (mo, O, E) bf16
     const int l4 = lane >> 4;
     const int we = wid >> 1;          // 0..3: e-group of 16
     const int wk = wid & 1;           // 0..1: k-group of 64
-    const int e0 = blockIdx.x * 64;
+    int eb, sp;
+    if (coh) {
+        int x = blockIdx.x & 7, r = blockIdx.x >> 3;
+        eb = r % nmemb;
+        sp = x + 8 * (r / nmemb);
+    } else {
+        eb = blockIdx.x % nmemb;
+        sp = blockIdx.x / nmemb;
+    }
+    const int e0 = eb * 64;
 
     f32x4 acc[4];                      // 16 e x 64 k per wave
 #pragma unroll
     for (int i = 0; i < 4; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
 
     const int nmo = mo / 8, nuc = miF / 32;
-    const int mb_lo = (nmo / nsplit) * blockIdx.y;
+    const int mb_lo = (nmo / nsplit) * sp;
     const int mb_hi = mb_lo + nmo / nsplit;
     for (int cb = 0; cb < nuc; ++cb) {
         // stage u chunk [32][O][64] once per urow-chunk
@@ -282,7 +291,7 @@ pairconv_bwd_du_kernel(const __bf16* __restrict__ H,   // (E,128)
                        const float* __restrict__ bias, // (mo*miF,)
                        const __bf16* __restrict__ Gt,  // (mo,O,E)
                        float* __restrict__ dU,         // (miF, O, E) f32
-                       int E, int mo, int miF) {
+                       int E, int mo, int miF, int nmemb, int coh) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
     __bf16* h_lds = reinterpret_cast<__bf16*>(smem);                   // [64][128] swizzled 16K
     __bf16* r_lds = reinterpret_cast<__bf16*>(smem + 16384);           // [256n][64e] 32K
@@ -298,7 +307,16 @@ pairconv_bwd_du_kernel(const __bf16* __restrict__ H,   // (E,128)
     const int l4 = lane >> 4;
     const int wm = wid >> 1;
     const int we = wid & 1;
-    const int e0 = blockIdx.x * 64;
+    int eb, sp;
+    if (coh) {
+        int x = blockIdx.x & 7, r = blockIdx.x >> 3;
+        eb = r % nmemb;
+        sp = x + 8 * (r / nmemb);
+    } else {
+        eb = blockIdx.x % nmemb;
+        sp = blockIdx.x / nmemb;
+    }
+    const int e0 = eb * 64;
     const int cb = blockIdx.y;        // urow chunk
     const int uc0 = cb * 32;
 
@@ -414,17 +432,18 @@ void pairconv_bwd_dh(torch::Tensor Gt, torch::Tensor Ut, torch::Tensor Wt,
     auto stream = at::cuda::getCurrentHIPStream();
     int eblk = (E + 63) / 64;
     int nmo = mo / 8;
-    int nsplit = 1;
+    int nsplit = (nmo % 8 == 0) ? 8 : 1;
     while (eblk * nsplit * 2 <= 1024 && nsplit * 2 <= nmo && nmo % (nsplit * 2) == 0)
         nsplit *= 2;
-    dim3 grid(eblk, nsplit);
+    int coh = (nsplit % 8 == 0) ? 1 : 0;
+    dim3 grid(eblk * nsplit);
     DISPATCH_O(O, {
         size_t lds = 32768 + (size_t)32 * kO * 64 * 2 + (size_t)8 * kO * 64 * 2;
         hipLaunchKernelGGL(HIP_KERNEL_NAME(pairconv_bwd_dh_kernel<kO>), grid, dim3(NT), lds, stream,
                            reinterpret_cast<const __bf16*>(Gt.data_ptr()),
                            reinterpret_cast<const __bf16*>(Ut.data_ptr()),
                            reinterpret_cast<const __bf16*>(Wt.data_ptr()),
-                           dH.data_ptr<float>(), E, mo, miF, nsplit);
+                           dH.data_ptr<float>(), E, mo, miF, nsplit, eblk, coh);
     });
     hipError_t err = hipGetLastError();
     TORCH_CHECK(err == hipSuccess, "bwd_dh: ", hipGetErrorString(err));
@@ -458,7 +477,10 @@ void pairconv_bwd_du(torch::Tensor H, torch::Tensor W, torch::Tensor bias,
     TORCH_CHECK(W.numel() == (int64_t)mo * miF * KDIM, "expect packed W (P)");
     TORCH_CHECK(bias.dtype() == torch::kFloat32);
     auto stream = at::cuda::getCurrentHIPStream();
-    dim3 grid((E + 63) / 64, miF / 32);
+    int nmemb = (E + 63) / 64;
+    int ncb = miF / 32;
+    int coh = (ncb % 8 == 0) ? 1 : 0;
+    dim3 grid(nmemb * ncb);
     DISPATCH_O(O, {
         size_t lds = 49152 + (size_t)((8 * kO * 64 * 2 + 15) & ~15) +
                      (size_t)32 * kO * 64 * 4 + 256 * 4;
@@ -467,7 +489,7 @@ void pairconv_bwd_du(torch::Tensor H, torch::Tensor W, torch::Tensor bias,
                            reinterpret_cast<const __bf16*>(W.data_ptr()),
                            bias.data_ptr<float>(),
                            reinterpret_cast<const __bf16*>(Gt.data_ptr()),
-                           dU.data_ptr<float>(), E, mo, miF);
+                           dU.data_ptr<float>(), E, mo, miF, nmemb, coh);
     });
     hipError_t err = hipGetLastError();
     TORCH_CHECK(err == hipSuccess, "bwd_du: ", hipGetErrorString(err));

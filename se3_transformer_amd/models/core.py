@@ -268,6 +268,7 @@ class ConvSE3(nn.Module):
                  edge_dim=0, fourier_encode_dist=False, num_fourier_features=4,
                  splits=4):
         super().__init__()
+        edge_dim = edge_dim if edge_dim is not None else 0
         self.fiber_in = fiber_in
         self.fiber_out = fiber_out
         self.edge_dim = edge_dim
