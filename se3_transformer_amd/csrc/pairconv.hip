@@ -65,17 +65,17 @@ pairconv_fwd_kernel(const __bf16* __restrict__ H,
     const int l15 = lane & 15;
     const int l4 = lane >> 4;          // 0..3
 
-    // cohort mapping: all concurrently-resident blocks on one XCD share the
-    // same mo-block (=> the same packed-W slice stays in that XCD's L2).
-    int eb, mb;
-    if (coh) {
-        int x = blockIdx.x & 7, r = blockIdx.x >> 3;
-        eb = r % nmemb;
-        mb = x + 8 * (r / nmemb);
-    } else {
-        eb = blockIdx.x % nmemb;
-        mb = blockIdx.x / nmemb;
-    }
+    // L3-panel mapping: order blocks in (PM mo-blocks x PE e-blocks) panels so
+    // the ~512 concurrently-resident blocks touch PM W-slices + PE u-slices,
+    // sized to stay inside the 256 MB Infinity Cache instead of re-streaming
+    // both operands from HBM per block.
+    const int PM = 16, PE = 32;
+    int panels_x = (nmemb + PE - 1) / PE;
+    int within = blockIdx.x % (PM * PE);
+    int panel = blockIdx.x / (PM * PE);
+    int eb = (panel % panels_x) * PE + within % PE;
+    int mb = (panel / panels_x) * PM + within / PE;
+    if (eb >= nmemb || mb >= coh) return;   // coh carries ng (mo/8)
     const int e0 = eb * BLK_E;
     const int mo0 = mb * BLK_MO;
 
@@ -227,8 +227,10 @@ static void launch_fwd(const torch::Tensor& H, const torch::Tensor& W,
                        int E, int mo, int miF) {
     int nmemb = (E + BLK_E - 1) / BLK_E;
     int ng = mo / BLK_MO;
-    int coh = (ng % 8 == 0) ? 1 : 0;
-    dim3 grid(nmemb * ng);
+    const int PM = 16, PE = 32;
+    int panels = ((nmemb + PE - 1) / PE) * ((ng + PM - 1) / PM);
+    int coh = ng;  // passed through as the mo-block count
+    dim3 grid((long)panels * PM * PE);
     size_t lds = 16384 + (size_t)UCHUNK * O * BLK_E * 2 + (size_t)BLK_E * BLK_MO * O * 4;
     auto stream = at::cuda::getCurrentHIPStream();
     hipLaunchKernelGGL(HIP_KERNEL_NAME(pairconv_fwd_kernel<O>), grid, dim3(NTHREADS), lds, stream,

@@ -56,15 +56,14 @@ pairconv_bwd_dh_kernel(const __bf16* __restrict__ Gt,  // (mo, O, E) bf16
     const int l4 = lane >> 4;
     const int we = wid >> 1;          // 0..3: e-group of 16
     const int wk = wid & 1;           // 0..1: k-group of 64
-    int eb, sp;
-    if (coh) {
-        int x = blockIdx.x & 7, r = blockIdx.x >> 3;
-        eb = r % nmemb;
-        sp = x + 8 * (r / nmemb);
-    } else {
-        eb = blockIdx.x % nmemb;
-        sp = blockIdx.x / nmemb;
-    }
+    // L3-panel mapping (see pairconv.hip): PS split-slices x PE e-blocks
+    const int PS = 4, PE = 128;
+    int panels_x = (nmemb + PE - 1) / PE;
+    int within = blockIdx.x % (PS * PE);
+    int panel = blockIdx.x / (PS * PE);
+    int eb = (panel % panels_x) * PE + within % PE;
+    int sp = (panel / panels_x) * PS + within / PE;
+    if (eb >= nmemb || sp >= nsplit) return;
     const int e0 = eb * 64;
 
     f32x4 acc[4];                      // 16 e x 64 k per wave
@@ -307,15 +306,14 @@ pairconv_bwd_du_kernel(const __bf16* __restrict__ H,   // (E,128)
     const int l4 = lane >> 4;
     const int wm = wid >> 1;
     const int we = wid & 1;
-    int eb, sp;
-    if (coh) {
-        int x = blockIdx.x & 7, r = blockIdx.x >> 3;
-        eb = r % nmemb;
-        sp = x + 8 * (r / nmemb);
-    } else {
-        eb = blockIdx.x % nmemb;
-        sp = blockIdx.x / nmemb;
-    }
+    // L3-panel mapping (see pairconv.hip): PS split-slices x PE e-blocks
+    const int PS = 4, PE = 128;
+    int panels_x = (nmemb + PE - 1) / PE;
+    int within = blockIdx.x % (PS * PE);
+    int panel = blockIdx.x / (PS * PE);
+    int eb = (panel % panels_x) * PE + within % PE;
+    int sp = (panel / panels_x) * PS + within / PE;
+    if (eb >= nmemb || sp >= nsplit) return;
     const int e0 = eb * 64;
     const int cb = blockIdx.y;        // urow chunk
     const int uc0 = cb * 32;
@@ -435,8 +433,10 @@ void pairconv_bwd_dh(torch::Tensor Gt, torch::Tensor Ut, torch::Tensor Wt,
     int nsplit = (nmo % 8 == 0) ? 8 : 1;
     while (eblk * nsplit * 2 <= 1024 && nsplit * 2 <= nmo && nmo % (nsplit * 2) == 0)
         nsplit *= 2;
-    int coh = (nsplit % 8 == 0) ? 1 : 0;
-    dim3 grid(eblk * nsplit);
+    int coh = 0;
+    const int PS = 4, PE = 128;
+    int panels = ((eblk + PE - 1) / PE) * ((nsplit + PS - 1) / PS);
+    dim3 grid((long)panels * PS * PE);
     DISPATCH_O(O, {
         size_t lds = 32768 + (size_t)32 * kO * 64 * 2 + (size_t)8 * kO * 64 * 2;
         hipLaunchKernelGGL(HIP_KERNEL_NAME(pairconv_bwd_dh_kernel<kO>), grid, dim3(NT), lds, stream,
@@ -479,8 +479,10 @@ void pairconv_bwd_du(torch::Tensor H, torch::Tensor W, torch::Tensor bias,
     auto stream = at::cuda::getCurrentHIPStream();
     int nmemb = (E + 63) / 64;
     int ncb = miF / 32;
-    int coh = (ncb % 8 == 0) ? 1 : 0;
-    dim3 grid(nmemb * ncb);
+    int coh = ncb;
+    const int PC = 16, PE = 32;
+    int panels = ((nmemb + PE - 1) / PE) * ((ncb + PC - 1) / PC);
+    dim3 grid((long)panels * PC * PE);
     DISPATCH_O(O, {
         size_t lds = 49152 + (size_t)((8 * kO * 64 * 2 + 15) & ~15) +
                      (size_t)32 * kO * 64 * 4 + 256 * 4;
