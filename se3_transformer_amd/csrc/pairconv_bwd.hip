@@ -306,17 +306,16 @@ pairconv_bwd_du_kernel(const __bf16* __restrict__ H,   // (E,128)
     const int l4 = lane >> 4;
     const int wm = wid >> 1;
     const int we = wid & 1;
-    // L3-panel mapping (see pairconv.hip): PS split-slices x PE e-blocks
-    const int PS = 4, PE = 128;
+    // L3-panel mapping (see pairconv.hip): PC urow-chunks x PE e-blocks
+    const int PC = 16, PE = 32;
     int panels_x = (nmemb + PE - 1) / PE;
-    int within = blockIdx.x % (PS * PE);
-    int panel = blockIdx.x / (PS * PE);
+    int within = blockIdx.x % (PC * PE);
+    int panel = blockIdx.x / (PC * PE);
     int eb = (panel % panels_x) * PE + within % PE;
-    int sp = (panel / panels_x) * PS + within / PE;
-    if (eb >= nmemb || sp >= nsplit) return;
+    int cb = (panel / panels_x) * PC + within / PE;
+    if (eb >= nmemb || cb >= coh) return;   // coh carries miF/32
     const int e0 = eb * 64;
-    const int cb = blockIdx.y;        // urow chunk
-    const int uc0 = cb * 32;
+    const int uc0 = cb * 32;          // urow chunk
 
     // stage H tile swizzled (as forward)
     for (int i = tid; i < (64 * KDIM) / 8; i += NT) {
