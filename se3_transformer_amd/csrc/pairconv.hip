@@ -270,6 +270,9 @@ void pairconv_fwd(torch::Tensor H, torch::Tensor W, torch::Tensor Ut,
 
 void sh_basis_fwd(torch::Tensor rel, torch::Tensor qcat, torch::Tensor normtab,
                   torch::Tensor meta, torch::Tensor out, int64_t L);
+void knn_graph(torch::Tensor coors, torch::Tensor nmask, torch::Tensor idx,
+               torch::Tensor dist, torch::Tensor rel, torch::Tensor m,
+               int64_t k, double radius, bool causal);
 void attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
               torch::Tensor mask, torch::Tensor out,
               int64_t n, int64_t heads, double scale);
@@ -289,6 +292,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("pairconv_bwd_dh", &pairconv_bwd_dh, "dH backward");
     m.def("pairconv_bwd_dw", &pairconv_bwd_dw, "dW backward");
     m.def("pairconv_bwd_du", &pairconv_bwd_du, "dU backward");
+    m.def("knn_graph", &knn_graph, "on-device kNN graph build");
     m.def("attn_fwd", &attn_fwd, "fused neighbor attention forward");
     m.def("norm_se3_fwd", &norm_se3_fwd, "fused NormSE3 forward");
     m.def("norm_se3_bwd", &norm_se3_bwd, "fused NormSE3 backward");

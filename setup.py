@@ -21,7 +21,8 @@ setup(
                      'se3_transformer_amd/csrc/pairconv_bwd.hip',
                      'se3_transformer_amd/csrc/sh_basis.hip',
                      'se3_transformer_amd/csrc/norm_se3.hip',
-                     'se3_transformer_amd/csrc/attn.hip'],
+                     'se3_transformer_amd/csrc/attn.hip',
+                     'se3_transformer_amd/csrc/knn.hip'],
             extra_compile_args={
                 'cxx': ['-O3'],
                 'nvcc': ['-O3', '--offload-arch=gfx950'],
