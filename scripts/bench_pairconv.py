@@ -30,6 +30,7 @@ def main():
     p.add_argument('--pair', default='3,3')
     p.add_argument('--E', type=int, default=9216)
     p.add_argument('--ch', type=int, default=512)
+    p.add_argument('--only', default=None, choices=['fwd','dh','dw','du'])
     args = p.parse_args()
     di, do = map(int, args.pair.split(','))
     F = 2 * min(di, do) + 1
@@ -53,6 +54,16 @@ def main():
 
     from se3_transformer_amd.ops.fused import _pack_w_dh, _pack_w_fwd
     P = _pack_w_fwd(W, mo, miF)
+    if args.only:
+        fn = {'fwd': lambda: _C.pairconv_fwd(H, P, Ut, out, mo),
+              'dh': lambda: _C.pairconv_bwd_dh(gt, Ut, _pack_w_dh(W, mo, miF),
+                                               torch.zeros(E, K, device=dev), mo),
+              'dw': lambda: _C.pairconv_bwd_dw(gt, Ut, H.t().contiguous(),
+                                               torch.empty(N, K, device=dev), mo),
+              'du': lambda: _C.pairconv_bwd_du(H, P, bias, gt,
+                                               torch.empty(miF, O, E, device=dev), mo)}[args.only]
+        print(args.only, timeit(fn, iters=3, warmup=1), 'ms')
+        return
     ms = timeit(lambda: _C.pairconv_fwd(H, P, Ut, out, mo))
     print(f'fwd    ({di},{do}): {ms:8.3f} ms  {gemm_fl/ms/1e9:7.1f} TF/s (gemm) '
           f'{(gemm_fl+epi_fl)/ms/1e9:7.1f} TF/s (total)')
