@@ -45,21 +45,17 @@ __device__ __forceinline__ float bf16_to_f32(unsigned short u) {
 }
 
 template <int O>
-__global__ void __launch_bounds__(NTHREADS, 2)   // 1 block/CU (LDS-bound anyway)
+__global__ void __launch_bounds__(NTHREADS, 4)   // cap VGPR<=128: 2 blocks/CU
 pairconv_fwd_kernel(const __bf16* __restrict__ H,
                     const __bf16* __restrict__ P,   // packed W: [mo/8][miF/32][wm4][mf4][kit4][lane64][8]
                     const __bf16* __restrict__ Ut,
                     float* __restrict__ out,
-                    int E, int mo, int miF, int nmemb, int ng2) {
-    // Supertile: each block owns 2 e-tiles x 2 mo-blocks (128 edges x 16 mo).
-    // Per chunk the W fragments are re-read for the 2nd e-tile from L2 and the
-    // u chunk is reused by both mo-blocks: HBM traffic per sample is half W
-    // + 1/16 u instead of (E/64) W re-streams (PMC: the kernel is W/u-stream
-    // bandwidth bound, SQ_INSTS_MFMA way below the pipe).
+                    int E, int mo, int miF, int nmemb, int coh) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
-    __bf16* h_lds = reinterpret_cast<__bf16*>(smem);                       // [2][64][128] 32 KiB
-    __bf16* u_lds = reinterpret_cast<__bf16*>(smem + 32768);               // [32][O][64]
-    float* part = reinterpret_cast<float*>(smem + 32768 + UCHUNK * O * BLK_E * 2); // [2et][2ms][64][8][O]
+    // carve: H tile | u chunk | partial accumulator
+    __bf16* h_lds = reinterpret_cast<__bf16*>(smem);                       // [64][128] swizzled, 16 KiB
+    __bf16* u_lds = reinterpret_cast<__bf16*>(smem + 16384);               // [32][O][64]
+    float* part = reinterpret_cast<float*>(smem + 16384 + UCHUNK * O * BLK_E * 2); // [64][8][O]
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -69,156 +65,158 @@ pairconv_fwd_kernel(const __bf16* __restrict__ H,
     const int l15 = lane & 15;
     const int l4 = lane >> 4;          // 0..3
 
-    const int eb = blockIdx.x % nmemb;      // 128-edge group
-    const int mb2 = blockIdx.x / nmemb;     // 16-mo group
-    const int e00 = eb * (2 * BLK_E);
-    const int mo00 = mb2 * (2 * BLK_MO);
-
-    // ---- stage both H tiles (2 x 64 x 128 bf16), XOR-swizzled 16B slots
-    for (int i = tid; i < (2 * BLK_E * KDIM) / 8; i += NTHREADS) {  // 16B units
-        int e = i >> 4;             // 0..127
-        int k16 = i & 15;
-        int dst = e * 256 + ((k16 * 16) ^ ((e & 15) << 4));
-        bf16x8 v;
-        if (e00 + e < E) {
-            v = *reinterpret_cast<const bf16x8*>(H + (size_t)(e00 + e) * KDIM + k16 * 8);
-        } else {
-            v = bf16x8(0);
-        }
-        *reinterpret_cast<bf16x8*>(reinterpret_cast<char*>(h_lds) + dst) = v;
+    // cohort mapping: all concurrently-resident blocks on one XCD share the
+    // same mo-block (=> the same packed-W slice stays in that XCD's L2).
+    int eb, mb;
+    if (coh) {
+        int x = blockIdx.x & 7, r = blockIdx.x >> 3;
+        eb = r % nmemb;
+        mb = x + 8 * (r / nmemb);
+    } else {
+        eb = blockIdx.x % nmemb;
+        mb = blockIdx.x / nmemb;
     }
-    for (int i = tid; i < 4 * BLK_E * BLK_MO * O; i += NTHREADS) part[i] = 0.f;
+    const int e0 = eb * BLK_E;
+    const int mo0 = mb * BLK_MO;
+
+    // ---- stage H tile (64 x 128 bf16), XOR-swizzled 16B slots within each row
+    {
+        for (int i = tid; i < (BLK_E * KDIM) / 8; i += NTHREADS) {  // 16B units
+            int e = i >> 4;             // 16 units per row
+            int k16 = i & 15;           // 16B slot
+            int dst = e * 256 + ((k16 * 16) ^ ((e & 15) << 4));
+            bf16x8 v;
+            if (e0 + e < E) {
+                v = *reinterpret_cast<const bf16x8*>(H + (size_t)(e0 + e) * KDIM + k16 * 8);
+            } else {
+                v = bf16x8(0);
+            }
+            *reinterpret_cast<bf16x8*>(reinterpret_cast<char*>(h_lds) + dst) = v;
+        }
+        // zero partial accumulator
+        for (int i = tid; i < BLK_E * BLK_MO * O; i += NTHREADS) part[i] = 0.f;
+    }
     __syncthreads();
 
     const int nchunks = miF / UCHUNK;
     for (int c = 0; c < nchunks; ++c) {
         const int uc0 = c * UCHUNK;
-#pragma unroll
-        for (int et = 0; et < 2; ++et) {
-            const int e0 = e00 + et * BLK_E;
-            // ---- stage u chunk for this e-tile: u_lds[urow][o][e]
-            for (int i = tid; i < (UCHUNK * O * BLK_E) / 8; i += NTHREADS) {
-                int ro = i >> 3;
-                int eu = (i & 7) * 8;
-                const __bf16* src = Ut + ((size_t)(uc0 + (ro / O)) * O + (ro % O)) * E + e0 + eu;
-                bf16x8 v;
-                if (e0 + eu + 8 <= E) {
-                    v = *reinterpret_cast<const bf16x8*>(src);
-                } else {
-                    for (int j = 0; j < 8; ++j)
-                        v[j] = (e0 + eu + j < E) ? src[j] : (__bf16)0.f;
-                }
-                *reinterpret_cast<bf16x8*>(u_lds + (size_t)ro * BLK_E + eu) = v;
+
+        // ---- stage u chunk: u_lds[urow][o][e] <- Ut[(uc0+urow)*O + o][e0..e0+64]
+        for (int i = tid; i < (UCHUNK * O * BLK_E) / 8; i += NTHREADS) { // 16B units
+            int ro = i >> 3;            // (urow*O + o)
+            int eu = (i & 7) * 8;       // e offset within 64
+            const __bf16* src = Ut + ((size_t)(uc0 + (ro / O)) * O + (ro % O)) * E + e0 + eu;
+            bf16x8 v;
+            if (e0 + eu + 8 <= E) {
+                v = *reinterpret_cast<const bf16x8*>(src);
+            } else {
+                for (int j = 0; j < 8; ++j)
+                    v[j] = (e0 + eu + j < E) ? src[j] : (__bf16)0.f;
             }
-            __syncthreads();
-
-#pragma unroll
-            for (int ms = 0; ms < 2; ++ms) {
-                // ---- GEMM: R^T tile (256 n x 64 e), K=128
-                f32x4 acc[4][2];
-#pragma unroll
-                for (int mf = 0; mf < 4; ++mf)
-#pragma unroll
-                    for (int ef = 0; ef < 2; ++ef) acc[mf][ef] = f32x4(0.f);
-
-                const __bf16* pbase = P
-                    + ((((size_t)(mb2 * 2 + ms) * (miF / 32) + c) * 4 + wm) * 4) * 4 * 64 * 8
-                    + (size_t)lane * 8;
-#pragma unroll
-                for (int kit = 0; kit < 4; ++kit) {
-                    const int k0 = kit * 32 + l4 * 8;
-                    bf16x8 a[4], b[2];
-#pragma unroll
-                    for (int mf = 0; mf < 4; ++mf) {
-                        a[mf] = *reinterpret_cast<const bf16x8*>(pbase + ((size_t)mf * 4 + kit) * 64 * 8);
-                    }
-#pragma unroll
-                    for (int ef = 0; ef < 2; ++ef) {
-                        int e = et * BLK_E + we * 32 + ef * 16 + l15;
-                        int byte = e * 256 + ((k0 * 2) ^ ((e & 15) << 4));
-                        b[ef] = *reinterpret_cast<const bf16x8*>(reinterpret_cast<char*>(h_lds) + byte);
-                    }
-#pragma unroll
-                    for (int mf = 0; mf < 4; ++mf)
-#pragma unroll
-                        for (int ef = 0; ef < 2; ++ef)
-                            acc[mf][ef] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[mf], b[ef], acc[mf][ef], 0, 0, 0);
-                }
-
-                // ---- epilogue: contract acc against u_lds into s[ef][moi][o]
-                float s[2][2][O];
-#pragma unroll
-                for (int ef = 0; ef < 2; ++ef)
-#pragma unroll
-                    for (int mi_ = 0; mi_ < 2; ++mi_)
-#pragma unroll
-                        for (int o = 0; o < O; ++o) s[ef][mi_][o] = 0.f;
-
-#pragma unroll
-                for (int mf = 0; mf < 2; ++mf) {   // mf and mf+2 share urow
-#pragma unroll
-                    for (int reg = 0; reg < 4; ++reg) {
-                        const int r = wm * 64 + mf * 16 + l4 * 4 + reg;
-                        const int urow = r & 31;
-#pragma unroll
-                        for (int ef = 0; ef < 2; ++ef) {
-                            const int e = we * 32 + ef * 16 + l15;
-                            const float rv0 = acc[mf][ef][reg];
-                            const float rv1 = acc[mf + 2][ef][reg];
-#pragma unroll
-                            for (int o = 0; o < O; ++o) {
-                                float uv = bf16_to_f32(
-                                    reinterpret_cast<const unsigned short*>(u_lds)[(urow * O + o) * BLK_E + e]);
-                                s[ef][0][o] = fmaf(rv0, uv, s[ef][0][o]);
-                                s[ef][1][o] = fmaf(rv1, uv, s[ef][1][o]);
-                            }
-                        }
-                    }
-                }
-
-#pragma unroll
-                for (int ef = 0; ef < 2; ++ef)
-#pragma unroll
-                    for (int mi_ = 0; mi_ < 2; ++mi_)
-#pragma unroll
-                        for (int o = 0; o < O; ++o) {
-                            float v = s[ef][mi_][o];
-                            v += __shfl_xor(v, 16);
-                            v += __shfl_xor(v, 32);
-                            s[ef][mi_][o] = v;
-                        }
-                float* pslab = part + (size_t)(et * 2 + ms) * BLK_E * BLK_MO * O;
-                if (l4 == 0) {
-#pragma unroll
-                    for (int ef = 0; ef < 2; ++ef) {
-                        const int e = we * 32 + ef * 16 + l15;
-#pragma unroll
-                        for (int mi_ = 0; mi_ < 2; ++mi_) {
-                            const int moi = wm * 2 + mi_;
-#pragma unroll
-                            for (int o = 0; o < O; ++o) {
-                                float* pp = pslab + ((size_t)e * BLK_MO + moi) * O + o;
-                                *pp += s[ef][mi_][o];
-                            }
-                        }
-                    }
-                }
-            }
-            __syncthreads();
+            *reinterpret_cast<bf16x8*>(u_lds + (size_t)ro * BLK_E + eu) = v;
         }
+        __syncthreads();
+
+        // ---- GEMM: R^T tile (256 n x 64 e), K=128
+        f32x4 acc[4][2];
+#pragma unroll
+        for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+            for (int ef = 0; ef < 2; ++ef) acc[mf][ef] = f32x4(0.f);
+
+        const __bf16* pbase = P + ((((size_t)mb * (miF / 32) + c) * 4 + wm) * 4) * 4 * 64 * 8
+                              + (size_t)lane * 8;
+#pragma unroll
+        for (int kit = 0; kit < 4; ++kit) {
+            const int k0 = kit * 32 + l4 * 8;
+            bf16x8 a[4], b[2];
+#pragma unroll
+            for (int mf = 0; mf < 4; ++mf) {
+                a[mf] = *reinterpret_cast<const bf16x8*>(pbase + ((size_t)mf * 4 + kit) * 64 * 8);
+            }
+#pragma unroll
+            for (int ef = 0; ef < 2; ++ef) {
+                int e = we * 32 + ef * 16 + l15;
+                int byte = e * 256 + ((k0 * 2) ^ ((e & 15) << 4));
+                b[ef] = *reinterpret_cast<const bf16x8*>(reinterpret_cast<char*>(h_lds) + byte);
+            }
+#pragma unroll
+            for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+                for (int ef = 0; ef < 2; ++ef)
+                    acc[mf][ef] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[mf], b[ef], acc[mf][ef], 0, 0, 0);
+        }
+
+        // ---- epilogue: contract acc against u_lds into s[ef][moi][o]
+        float s[2][2][O];
+#pragma unroll
+        for (int ef = 0; ef < 2; ++ef)
+#pragma unroll
+            for (int mi_ = 0; mi_ < 2; ++mi_)
+#pragma unroll
+                for (int o = 0; o < O; ++o) s[ef][mi_][o] = 0.f;
+
+#pragma unroll
+        for (int mf = 0; mf < 2; ++mf) {   // mf and mf+2 share urow (rows r, r+32)
+#pragma unroll
+            for (int reg = 0; reg < 4; ++reg) {
+                const int r = wm * 64 + mf * 16 + l4 * 4 + reg;
+                const int urow = r & 31;
+#pragma unroll
+                for (int ef = 0; ef < 2; ++ef) {
+                    const int e = we * 32 + ef * 16 + l15;
+                    const float rv0 = acc[mf][ef][reg];
+                    const float rv1 = acc[mf + 2][ef][reg];
+#pragma unroll
+                    for (int o = 0; o < O; ++o) {
+                        float uv = bf16_to_f32(
+                            reinterpret_cast<const unsigned short*>(u_lds)[(urow * O + o) * BLK_E + e]);
+                        s[ef][0][o] = fmaf(rv0, uv, s[ef][0][o]);
+                        s[ef][1][o] = fmaf(rv1, uv, s[ef][1][o]);
+                    }
+                }
+            }
+        }
+
+        // cross-lane reduce over l4 groups (rows), then accumulate into LDS partial
+#pragma unroll
+        for (int ef = 0; ef < 2; ++ef)
+#pragma unroll
+            for (int mi_ = 0; mi_ < 2; ++mi_)
+#pragma unroll
+                for (int o = 0; o < O; ++o) {
+                    float v = s[ef][mi_][o];
+                    v += __shfl_xor(v, 16);
+                    v += __shfl_xor(v, 32);
+                    s[ef][mi_][o] = v;
+                }
+        if (l4 == 0) {
+#pragma unroll
+            for (int ef = 0; ef < 2; ++ef) {
+                const int e = we * 32 + ef * 16 + l15;
+#pragma unroll
+                for (int mi_ = 0; mi_ < 2; ++mi_) {
+                    const int moi = wm * 2 + mi_;
+#pragma unroll
+                    for (int o = 0; o < O; ++o) {
+                        float* p = part + ((size_t)e * BLK_MO + moi) * O + o;
+                        *p += s[ef][mi_][o];
+                    }
+                }
+            }
+        }
+        __syncthreads();
     }
 
-    // ---- write out: out[e00 + et*64 + e][mo00 + ms*8 + moi][o] += partial
-    for (int i = tid; i < 4 * BLK_E * BLK_MO * O; i += NTHREADS) {
+    // ---- write out: out[e0+e][mo0+moi][o] += partial
+    for (int i = tid; i < BLK_E * BLK_MO * O; i += NTHREADS) {
         int o = i % O;
         int moi = (i / O) % BLK_MO;
-        int e = (i / (O * BLK_MO)) % BLK_E;
-        int slab = i / (O * BLK_MO * BLK_E);   // et*2 + ms
-        int et = slab >> 1, ms = slab & 1;
-        int eg = e00 + et * BLK_E + e;
-        if (eg < E) {
-            float* pp = out + ((size_t)eg * mo + mo00 + ms * BLK_MO + moi) * O + o;
-            *pp += part[i];
+        int e = i / (O * BLK_MO);
+        if (e0 + e < E) {
+            float* p = out + ((size_t)(e0 + e) * mo + mo0 + moi) * O + o;
+            *p += part[((size_t)e * BLK_MO + moi) * O + o];
         }
     }
 }
@@ -227,16 +225,17 @@ template <int O>
 static void launch_fwd(const torch::Tensor& H, const torch::Tensor& W,
                        const torch::Tensor& Ut, torch::Tensor& out,
                        int E, int mo, int miF) {
-    int nmemb = (E + 2 * BLK_E - 1) / (2 * BLK_E);
-    int ng2 = mo / (2 * BLK_MO);
-    dim3 grid((long)nmemb * ng2);
-    size_t lds = 32768 + (size_t)UCHUNK * O * BLK_E * 2 + (size_t)4 * BLK_E * BLK_MO * O * 4;
+    int nmemb = (E + BLK_E - 1) / BLK_E;
+    int ng = mo / BLK_MO;
+    int coh = (ng % 8 == 0) ? 1 : 0;
+    dim3 grid(nmemb * ng);
+    size_t lds = 16384 + (size_t)UCHUNK * O * BLK_E * 2 + (size_t)BLK_E * BLK_MO * O * 4;
     auto stream = at::cuda::getCurrentHIPStream();
     hipLaunchKernelGGL(HIP_KERNEL_NAME(pairconv_fwd_kernel<O>), grid, dim3(NTHREADS), lds, stream,
                        reinterpret_cast<const __bf16*>(H.data_ptr()),
                        reinterpret_cast<const __bf16*>(W.data_ptr()),
                        reinterpret_cast<const __bf16*>(Ut.data_ptr()),
-                       out.data_ptr<float>(), E, mo, miF, nmemb, ng2);  // W arg = packed P
+                       out.data_ptr<float>(), E, mo, miF, nmemb, coh);  // W arg = packed P
 }
 
 void pairconv_fwd(torch::Tensor H, torch::Tensor W, torch::Tensor Ut,
@@ -255,7 +254,7 @@ void pairconv_fwd(torch::Tensor H, torch::Tensor W, torch::Tensor Ut,
     TORCH_CHECK(Ut.size(2) == E);
     TORCH_CHECK(out.size(0) == E && out.size(1) == mo && out.size(2) == O);
     TORCH_CHECK(miF % UCHUNK == 0, "miF must be a multiple of 32");
-    TORCH_CHECK(mo % (2 * BLK_MO) == 0, "mo must be a multiple of 16");
+    TORCH_CHECK(mo % BLK_MO == 0, "mo must be a multiple of 8");
     switch (O) {
         case 1: launch_fwd<1>(H, W, Ut, out, E, mo, miF); break;
         case 3: launch_fwd<3>(H, W, Ut, out, E, mo, miF); break;

@@ -45,7 +45,7 @@ def require_ext():
 
 
 def fused_shapes_ok(mo: int, miF: int, O: int, mid_dim: int) -> bool:
-    return mid_dim == 128 and miF % 32 == 0 and mo % 16 == 0 and O in (1, 3, 5, 7)
+    return mid_dim == 128 and miF % 32 == 0 and mo % 8 == 0 and O in (1, 3, 5, 7)
 
 
 def _pack_w_fwd(W16, mo, miF):
