@@ -107,16 +107,28 @@ def main():
 
     # hipGraph capture: the step is shape-static, so capture once and replay —
     # removes the Python/launch-gap overhead between the ~10k kernels per step.
-    graph_mode = (args.graph if args.graph is not None else world == 1) \
-        and use_cuda and world == 1
+    # For world > 1 the RCCL all-reduces are captured too (stream-ordered);
+    # if capture fails for any reason we fall back to the eager step.
+    graph_mode = (args.graph if args.graph is not None else True) and use_cuda
     if graph_mode:
-        for _ in range(2):          # allocator warmup before capture
-            train_step()
-        torch.cuda.synchronize()
-        g_step = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(g_step):
-            train_step()
-        train_step = lambda: g_step.replay()  # noqa: E731
+        try:
+            for _ in range(2):      # allocator warmup before capture
+                train_step()
+            torch.cuda.synchronize()
+            if world > 1:
+                torch.distributed.barrier()
+            g_step = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g_step):
+                train_step()
+            train_step = lambda: g_step.replay()  # noqa: E731
+            if rank == 0:
+                print('[bench] hipGraph capture ok', file=sys.stderr, flush=True)
+        except Exception as e:      # pragma: no cover
+            if rank == 0:
+                print(f'[bench] graph capture failed ({e}); eager steps',
+                      file=sys.stderr, flush=True)
+
+    import sys
 
     def barrier_sync():
         if world > 1:
@@ -124,7 +136,6 @@ def main():
         if use_cuda:
             torch.cuda.synchronize()
 
-    import sys
     for i in range(args.warmup):
         t = time.perf_counter()
         train_step()
