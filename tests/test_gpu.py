@@ -61,3 +61,47 @@ def test_bf16_autocast_finite():
     out.float().pow(2).mean().backward()
     torch.cuda.synchronize()
     assert torch.isfinite(out).all()
+
+
+@needs_gpu
+def test_reversible_bf16_fused():
+    model = SE3Transformer(dim=32, heads=2, dim_head=16, depth=2,
+                           num_degrees=2, num_neighbors=6,
+                           reversible=True).to('cuda')
+    feats = torch.randn(1, 48, 32, device='cuda')
+    coors = torch.randn(1, 48, 3, device='cuda')
+    mask = torch.ones(1, 48, dtype=torch.bool, device='cuda')
+    with torch.autocast(device_type='cuda', dtype=torch.bfloat16):
+        out = model(feats, coors, mask, return_type=0)
+    out.float().pow(2).mean().backward()
+    torch.cuda.synchronize()
+    assert all(torch.isfinite(p.grad).all() for p in model.parameters()
+               if p.grad is not None)
+
+
+@needs_gpu
+def test_egnn_trunk_cuda():
+    model = SE3Transformer(dim=32, depth=2, num_degrees=2, num_neighbors=6,
+                           use_egnn=True, egnn_hidden_dim=16,
+                           num_edge_tokens=4, edge_dim=4).to('cuda')
+    feats = torch.randn(1, 48, 32, device='cuda')
+    coors = torch.randn(1, 48, 3, device='cuda')
+    mask = torch.ones(1, 48, dtype=torch.bool, device='cuda')
+    edges = torch.randint(0, 4, (1, 48, 48), device='cuda')
+    out = model(feats, coors, mask, edges=edges, return_type=0)
+    out.pow(2).mean().backward()
+    torch.cuda.synchronize()
+    assert torch.isfinite(out).all()
+
+
+@needs_gpu
+def test_one_headed_kv_cuda_bf16():
+    model = SE3Transformer(dim=32, heads=2, dim_head=16, depth=1,
+                           num_degrees=2, num_neighbors=6,
+                           one_headed_key_values=True).to('cuda')
+    feats = torch.randn(1, 48, 32, device='cuda')
+    coors = torch.randn(1, 48, 3, device='cuda')
+    mask = torch.ones(1, 48, dtype=torch.bool, device='cuda')
+    with torch.autocast(device_type='cuda', dtype=torch.bfloat16):
+        out = model(feats, coors, mask, return_type=0)
+    assert torch.isfinite(out).all()
