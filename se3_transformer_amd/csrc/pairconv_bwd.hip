@@ -306,14 +306,17 @@ pairconv_bwd_du_kernel(const __bf16* __restrict__ H,   // (E,128)
     const int l4 = lane >> 4;
     const int wm = wid >> 1;
     const int we = wid & 1;
-    // L3-panel mapping (see pairconv.hip): PC urow-chunks x PE e-blocks
-    const int PC = 16, PE = 32;
-    int panels_x = (nmemb + PE - 1) / PE;
-    int within = blockIdx.x % (PC * PE);
-    int panel = blockIdx.x / (PC * PE);
-    int eb = (panel % panels_x) * PE + within % PE;
-    int cb = (panel / panels_x) * PC + within / PE;
-    if (eb >= nmemb || cb >= coh) return;   // coh carries miF/32
+    // cohort mapping: concurrently-resident blocks on one XCD share a urow
+    // chunk (the same packed-W column slice stays L2-resident).
+    int eb, cb;
+    if (coh) {
+        int x = blockIdx.x & 7, r = blockIdx.x >> 3;
+        eb = r % nmemb;
+        cb = x + 8 * (r / nmemb);
+    } else {
+        eb = blockIdx.x % nmemb;
+        cb = blockIdx.x / nmemb;
+    }
     const int e0 = eb * 64;
     const int uc0 = cb * 32;          // urow chunk
 
@@ -429,7 +432,7 @@ void pairconv_bwd_dh(torch::Tensor Gt, torch::Tensor Ut, torch::Tensor Wt,
     auto stream = at::cuda::getCurrentHIPStream();
     int eblk = (E + 63) / 64;
     int nmo = mo / 8;
-    int nsplit = (nmo % 8 == 0) ? 8 : 1;
+    int nsplit = (nmo % 16 == 0) ? 16 : ((nmo % 8 == 0) ? 8 : 1);
     while (eblk * nsplit * 2 <= 1024 && nsplit * 2 <= nmo && nmo % (nsplit * 2) == 0)
         nsplit *= 2;
     int coh = 0;
@@ -478,10 +481,8 @@ void pairconv_bwd_du(torch::Tensor H, torch::Tensor W, torch::Tensor bias,
     auto stream = at::cuda::getCurrentHIPStream();
     int nmemb = (E + 63) / 64;
     int ncb = miF / 32;
-    int coh = ncb;
-    const int PC = 16, PE = 32;
-    int panels = ((nmemb + PE - 1) / PE) * ((ncb + PC - 1) / PC);
-    dim3 grid((long)panels * PC * PE);
+    int coh = (ncb % 8 == 0) ? 1 : 0;
+    dim3 grid(nmemb * ncb);
     DISPATCH_O(O, {
         size_t lds = 49152 + (size_t)((8 * kO * 64 * 2 + 15) & ~15) +
                      (size_t)32 * kO * 64 * 4 + 256 * 4;
