@@ -16,6 +16,7 @@ from __future__ import annotations
 import argparse
 import json
 import os
+import sys
 import time
 
 import torch
@@ -127,8 +128,6 @@ def main():
             if rank == 0:
                 print(f'[bench] graph capture failed ({e}); eager steps',
                       file=sys.stderr, flush=True)
-
-    import sys
 
     def barrier_sync():
         if world > 1:
