@@ -60,6 +60,8 @@ def main():
                         '(default: on for single-GPU CUDA runs)')
     p.add_argument('--no-graph', dest='graph', action='store_false')
     p.add_argument('--device', type=str, default=None)
+    p.add_argument('--timers', action='store_true',
+                   help='report per-phase (fwd/bwd/opt) hipEvent timings')
     args = p.parse_args()
 
     rank, world, local_rank = setup_distributed()
@@ -91,20 +93,42 @@ def main():
     autocast_dtype = torch.bfloat16 if args.dtype == 'bf16' else torch.float32
     autocast_enabled = args.dtype == 'bf16'
 
+    phase_events = []
+    if args.timers and use_cuda:
+        phase_events = [torch.cuda.Event(enable_timing=True) for _ in range(4)]
+
     def train_step():
         if ddp is not None:
             ddp.zero_grad_buffers()
         else:
             opt.zero_grad(set_to_none=False)
+        if phase_events:
+            phase_events[0].record()
         with torch.autocast(device_type=device.type, dtype=autocast_dtype,
                             enabled=autocast_enabled):
             out = runner(feats, coors, mask, return_type=0)
             loss = (out.float() - target).pow(2).mean()
+        if phase_events:
+            phase_events[1].record()
         loss.backward()
         if ddp is not None:
             ddp.finalize()
+        if phase_events:
+            phase_events[2].record()
         opt.step()
+        if phase_events:
+            phase_events[3].record()
         return loss
+
+    def report_phases(tag):
+        if not phase_events:
+            return
+        torch.cuda.synchronize()
+        f = phase_events[0].elapsed_time(phase_events[1])
+        bw = phase_events[1].elapsed_time(phase_events[2])
+        op = phase_events[2].elapsed_time(phase_events[3])
+        print(f'[bench] {tag}: fwd {f:.1f} ms | bwd {bw:.1f} ms | opt {op:.1f} ms',
+              file=sys.stderr, flush=True)
 
     # hipGraph capture: the step is shape-static, so capture once and replay —
     # removes the Python/launch-gap overhead between the ~10k kernels per step.
@@ -143,6 +167,7 @@ def main():
         if rank == 0:
             print(f'[bench] warmup {i}: {time.perf_counter() - t:.2f}s',
                   file=sys.stderr, flush=True)
+            report_phases(f'warmup {i}')
 
     barrier_sync()
     t0 = time.perf_counter()

@@ -9,11 +9,17 @@ import os
 
 os.environ.setdefault('PYTORCH_ROCM_ARCH', 'gfx950')
 
-from setuptools import setup
+from setuptools import find_packages, setup
 from torch.utils.cpp_extension import BuildExtension, CUDAExtension
 
 setup(
-    name='se3_transformer_amd_ext',
+    name='se3_transformer_amd',
+    version='0.9.0',
+    description='MI355X-native SE(3)-equivariant transformer '
+                '(API-compatible with se3-transformer-pytorch 0.9.0)',
+    packages=find_packages(include=['se3_transformer_amd*']),
+    python_requires='>=3.9',
+    install_requires=['torch>=2.0', 'numpy'],
     ext_modules=[
         CUDAExtension(
             name='se3_transformer_amd._C',

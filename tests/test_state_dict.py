@@ -118,3 +118,25 @@ def test_rotary_inv_freq_buffer():
                            rotary_position=True, rotary_rel_dist=True)
     assert 'rotary_pos_emb.inv_freq' in _names(model)
     assert model.state_dict()['rotary_pos_emb.inv_freq'].shape == (2,)
+
+
+def test_checkpoint_roundtrip(tmp_path):
+    import torch
+    from se3_transformer_amd import SE3Transformer
+    torch.manual_seed(0)
+    m1 = SE3Transformer(dim=16, heads=2, dim_head=8, depth=1, num_degrees=2,
+                        num_neighbors=4)
+    feats = torch.randn(1, 12, 16)
+    coors = torch.randn(1, 12, 3)
+    mask = torch.ones(1, 12, dtype=torch.bool)
+    ref = m1(feats, coors, mask, return_type=0)
+    torch.save(m1.state_dict(), tmp_path / 'ckpt.pt')
+
+    torch.manual_seed(123)
+    m2 = SE3Transformer(dim=16, heads=2, dim_head=8, depth=1, num_degrees=2,
+                        num_neighbors=4)
+    missing, unexpected = m2.load_state_dict(
+        torch.load(tmp_path / 'ckpt.pt', weights_only=True))
+    assert not missing and not unexpected
+    out = m2(feats, coors, mask, return_type=0)
+    assert torch.allclose(out, ref, atol=1e-6)
