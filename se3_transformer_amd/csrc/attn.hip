@@ -23,7 +23,7 @@ __global__ void __launch_bounds__(NTA)
 attn_fwd_kernel(const T* __restrict__ q, const T* __restrict__ k,
                 const T* __restrict__ v, const unsigned char* __restrict__ mask,
                 float* __restrict__ out, long R, int J, int DM, int n,
-                int heads, float scale) {
+                int heads, float scale, int kv_one) {
     const int lane = threadIdx.x & 63;
     const long r = ((long)blockIdx.x * (NTA / 64)) + (threadIdx.x >> 6);
     if (r >= R) return;
@@ -35,6 +35,11 @@ attn_fwd_kernel(const T* __restrict__ q, const T* __restrict__ k,
         int d = t * 64 + lane;
         qv[t] = (d < DM) ? (float)q[r * DM + d] : 0.f;
     }
+
+    // one-headed KV: all heads of a (b, i) row share one key/value row
+    long b0_ = r / ((long)heads * n);
+    long i0_ = r % n;
+    const long kvr = kv_one ? (b0_ * n + i0_) : r;
 
     // logits: lane j holds logit_j
     float logit = -3.0e38f;
@@ -50,7 +55,7 @@ attn_fwd_kernel(const T* __restrict__ q, const T* __restrict__ k,
 #pragma unroll
         for (int t = 0; t < DT; ++t) {
             int d = t * 64 + lane;
-            part += qv[t] * ((d < DM) ? (float)k[(r * J + j) * DM + d] : 0.f);
+            part += qv[t] * ((d < DM) ? (float)k[(kvr * J + j) * DM + d] : 0.f);
         }
 #pragma unroll
         for (int off = 32; off > 0; off >>= 1)
@@ -84,7 +89,7 @@ attn_fwd_kernel(const T* __restrict__ q, const T* __restrict__ k,
         for (int t = 0; t < DT; ++t) {
             int d = t * 64 + lane;
             if (d < DM)
-                ov[t] = fmaf(aj, (float)v[(r * J + j) * DM + d], ov[t]);
+                ov[t] = fmaf(aj, (float)v[(kvr * J + j) * DM + d], ov[t]);
         }
     }
 #pragma unroll
@@ -103,7 +108,7 @@ attn_fwd_kernel(const T* __restrict__ q, const T* __restrict__ k,
 
 void attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
               torch::Tensor mask, torch::Tensor out,
-              int64_t n, int64_t heads, double scale) {
+              int64_t n, int64_t heads, double scale, bool kv_one) {
     TORCH_CHECK(q.is_cuda() && q.is_contiguous() && k.is_contiguous() &&
                 v.is_contiguous() && out.is_contiguous());
     long R = q.size(0);
@@ -124,7 +129,8 @@ void attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                                grid, dim3(NTA), 0, stream,
                                q.data_ptr<float>(), k.data_ptr<float>(),
                                v.data_ptr<float>(), mptr, out.data_ptr<float>(),
-                               R, J, DM, (int)n, (int)heads, (float)scale);
+                               R, J, DM, (int)n, (int)heads, (float)scale,
+                               kv_one ? 1 : 0);
         } else {
             TORCH_CHECK(q.dtype() == torch::kBFloat16);
             hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_fwd_kernel<__hip_bfloat16, kDT>),
@@ -133,7 +139,8 @@ void attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                                reinterpret_cast<const __hip_bfloat16*>(k.data_ptr()),
                                reinterpret_cast<const __hip_bfloat16*>(v.data_ptr()),
                                mptr, out.data_ptr<float>(),
-                               R, J, DM, (int)n, (int)heads, (float)scale);
+                               R, J, DM, (int)n, (int)heads, (float)scale,
+                               kv_one ? 1 : 0);
         }
     });
     hipError_t err = hipGetLastError();

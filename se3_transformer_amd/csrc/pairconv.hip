@@ -273,7 +273,7 @@ void knn_graph(torch::Tensor coors, torch::Tensor nmask, torch::Tensor idx,
                int64_t k, double radius, bool causal);
 void attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
               torch::Tensor mask, torch::Tensor out,
-              int64_t n, int64_t heads, double scale);
+              int64_t n, int64_t heads, double scale, bool kv_one);
 void norm_se3_fwd(torch::Tensor t, torch::Tensor scale, torch::Tensor out, double eps);
 void norm_se3_bwd(torch::Tensor t, torch::Tensor scale, torch::Tensor dout,
                   torch::Tensor dt, torch::Tensor dscale, double eps);

@@ -301,15 +301,34 @@ class OneHeadedKVAttentionSE3(nn.Module):
                 k = torch.cat((global_k, k), dim=2)
                 v = torch.cat((global_v, v), dim=2)
 
-            sim = torch.einsum('bhidm,bijdm->bhij', q, k) * self.scale
+            bq, hq, nq, dq, mq = q.shape
+            J = k.shape[2]
+            from ..ops import fused as _fused
+            if (q.is_cuda and J <= 64 and dq * mq <= 448
+                    and q.dtype in (torch.float32, torch.bfloat16)
+                    and os.environ.get('SE3_EAGER_ATTN') != '1'
+                    and _fused.ext_available()):
+                mask_u8 = None
+                if neighbor_mask is not None:
+                    pad = J - neighbor_mask.shape[-1]
+                    mask_u8 = F.pad(neighbor_mask, (pad, 0), value=True) \
+                        .squeeze(1).to(torch.uint8).contiguous()
+                out = _fused.fused_attention(
+                    q.reshape(bq * hq * nq, dq * mq).contiguous(),
+                    k.to(q.dtype).reshape(bq * nq, J, dq * mq).contiguous(),
+                    v.to(q.dtype).reshape(bq * nq, J, dq * mq).contiguous(),
+                    mask_u8, nq, hq, self.scale, True)
+                out = out.view(bq, hq, nq, dq, mq).to(q.dtype)
+            else:
+                sim = torch.einsum('bhidm,bijdm->bhij', q, k) * self.scale
 
-            if neighbor_mask is not None:
-                num_left_pad = sim.shape[-1] - neighbor_mask.shape[-1]
-                padded_mask = F.pad(neighbor_mask, (num_left_pad, 0), value=True)
-                sim = sim.masked_fill(~padded_mask, -torch.finfo(sim.dtype).max)
+                if neighbor_mask is not None:
+                    num_left_pad = sim.shape[-1] - neighbor_mask.shape[-1]
+                    padded_mask = F.pad(neighbor_mask, (num_left_pad, 0), value=True)
+                    sim = sim.masked_fill(~padded_mask, -torch.finfo(sim.dtype).max)
 
-            attn = sim.softmax(dim=-1)
-            out = torch.einsum('bhij,bijdm->bhidm', attn, v)
+                attn = sim.softmax(dim=-1)
+                out = torch.einsum('bhij,bijdm->bhidm', attn, v)
             outputs[degree] = out.permute(0, 2, 1, 3, 4).reshape(b, n, -1, m)
 
         return self.to_out(outputs)

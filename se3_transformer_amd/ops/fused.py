@@ -206,27 +206,48 @@ class _AttnFn(torch.autograd.Function):
     Backward is the standard softmax-attention adjoint in library ops."""
 
     @staticmethod
-    def forward(ctx, q, k, v, mask_u8, n, heads, scale):
+    def forward(ctx, q, k, v, mask_u8, n, heads, scale, kv_one=False):
         ext = _load_ext()
         R, DM = q.shape
         out = torch.empty(R, DM, dtype=torch.float32, device=q.device)
         m = mask_u8 if mask_u8 is not None else \
             torch.empty(0, dtype=torch.uint8, device=q.device)
-        ext.attn_fwd(q, k, v, m, out, n, heads, scale)
+        ext.attn_fwd(q, k, v, m, out, n, heads, scale, kv_one)
         ctx.save_for_backward(q, k, v,
                               mask_u8 if mask_u8 is not None else None)
-        ctx.dims = (n, heads, scale)
+        ctx.dims = (n, heads, scale, kv_one)
         return out
 
     @staticmethod
     def backward(ctx, g):
         q, k, v, mask_u8 = ctx.saved_tensors
-        n, heads, scale = ctx.dims
-        R, J, DM = k.shape
-        qf, kf, vf, gf = q.float(), k.float(), v.float(), g.contiguous().float()
+        n, heads, scale, kv_one = ctx.dims
+        Rkv, J, DM = k.shape
+        R = q.shape[0]
+        b = R // (heads * n)
+        qf, gf = q.float(), g.contiguous().float()
+        if kv_one:
+            kf = k.float().view(b, 1, n, J, DM)
+            vf = v.float().view(b, 1, n, J, DM)
+            q4 = qf.view(b, heads, n, DM)
+            g4 = gf.view(b, heads, n, DM)
+            sim = torch.einsum('bhnd,bxnjd->bhnj', q4, kf) * scale
+            if mask_u8 is not None:
+                mrows = mask_u8.view(b, 1, n, J).bool()
+                sim = sim.masked_fill(~mrows, -torch.finfo(sim.dtype).max)
+            attn = sim.softmax(dim=-1)
+            dv = torch.einsum('bhnj,bhnd->bnjd', attn, g4)
+            dattn = torch.einsum('bhnd,bxnjd->bhnj', g4, vf)
+            dsim = attn * (dattn - (attn * dattn).sum(-1, keepdim=True))
+            dq = torch.einsum('bhnj,bxnjd->bhnd', dsim, kf) * scale
+            dk = torch.einsum('bhnj,bhnd->bnjd', dsim, q4) * scale
+            return (dq.reshape(R, DM).to(q.dtype),
+                    dk.reshape(Rkv, J, DM).to(k.dtype),
+                    dv.reshape(Rkv, J, DM).to(v.dtype),
+                    None, None, None, None, None)
+        kf, vf = k.float(), v.float()
         sim = torch.einsum('rd,rjd->rj', qf, kf) * scale
         if mask_u8 is not None:
-            b = R // (heads * n)
             mrows = mask_u8.view(b, 1, n, J).expand(b, heads, n, J) \
                 .reshape(R, J).bool()
             sim = sim.masked_fill(~mrows, -torch.finfo(sim.dtype).max)
@@ -237,8 +258,8 @@ class _AttnFn(torch.autograd.Function):
         dq = torch.einsum('rj,rjd->rd', dsim, kf) * scale
         dk = torch.einsum('rj,rd->rjd', dsim, qf) * scale
         return (dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype),
-                None, None, None, None)
+                None, None, None, None, None)
 
 
-def fused_attention(q, k, v, mask_u8, n, heads, scale):
-    return _AttnFn.apply(q, k, v, mask_u8, n, heads, scale)
+def fused_attention(q, k, v, mask_u8, n, heads, scale, kv_one=False):
+    return _AttnFn.apply(q, k, v, mask_u8, n, heads, scale, kv_one)

@@ -257,3 +257,38 @@ def test_knn_kernel_vs_eager():
     out = model(feats, coors, mask, return_type=0)
     err = (out - ref).abs().max().item()
     assert err < 1e-4, f'knn path mismatch: {err}'
+
+
+@needs_gpu
+def test_fused_attention_one_headed_vs_eager():
+    import os as _os
+    from se3_transformer_amd.models.attention import OneHeadedKVAttentionSE3
+    from se3_transformer_amd.models.fiber import Fiber
+    from se3_transformer_amd.ops.basis import get_basis_packed
+    torch.manual_seed(7)
+    fiber = Fiber([(0, 32), (1, 32)])
+    attn = OneHeadedKVAttentionSE3(fiber, dim_head=16, heads=2,
+                                   attend_self=True).to('cuda')
+    b, n, kn = 2, 36, 5
+    feats = {'0': torch.randn(b, n, 32, 1, device='cuda', requires_grad=True),
+             '1': torch.randn(b, n, 32, 3, device='cuda', requires_grad=True)}
+    feats2 = {k: v.detach().clone().requires_grad_(True) for k, v in feats.items()}
+    edge_info = (torch.randint(0, n, (b, n, kn), device='cuda'),
+                 torch.rand(b, n, kn, device='cuda') > 0.2, None)
+    rel_dist = torch.rand(b, n, kn, device='cuda')
+    basis = get_basis_packed(torch.randn(b, n, kn, 3, device='cuda'), 1)
+
+    _os.environ['SE3_EAGER_ATTN'] = '1'
+    try:
+        ref = attn(feats, edge_info, rel_dist, basis)
+    finally:
+        del _os.environ['SE3_EAGER_ATTN']
+    (ref['0'].pow(2).mean() + ref['1'].pow(2).mean()).backward()
+    gref = {k: v.grad.clone() for k, v in feats.items()}
+
+    out = attn(feats2, edge_info, rel_dist, basis)
+    for d in ref:
+        assert _rel_err(out[d], ref[d]) < 1e-4, d
+    (out['0'].pow(2).mean() + out['1'].pow(2).mean()).backward()
+    for d in gref:
+        assert _rel_err(feats2[d].grad, gref[d]) < 1e-3, d
