@@ -78,7 +78,7 @@ def main():
     model = build_model(args, device)
     n_params = sum(p_.numel() for p_ in model.parameters())
 
-    ddp = DistributedDataParallelSE3(model) if world > 1 else None
+    ddp = DistributedDataParallelSE3(model, sync_params=False) if world > 1 else None
     runner = ddp if ddp is not None else model
 
     opt = torch.optim.SGD(model.parameters(), lr=1e-4)

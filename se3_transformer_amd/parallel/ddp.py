@@ -67,10 +67,11 @@ class DistributedDataParallelSE3(nn.Module):
 
     def __init__(self, module: nn.Module, bucket_bytes: int = 128 << 20,
                  process_group=None, average: bool = True,
-                 grad_compression: str = 'none'):
+                 grad_compression: str = 'none', sync_params: bool = True):
         """grad_compression='bf16' all-reduces a bf16 copy of each bucket
         (halves xGMI traffic; fp32 master grads are restored from the reduced
-        bf16 values)."""
+        bf16 values). sync_params=False skips the initial parameter broadcast
+        (for ranks that already hold identical weights, e.g. seeded init)."""
         super().__init__()
         assert grad_compression in ('none', 'bf16')
         self.module = module
@@ -84,7 +85,7 @@ class DistributedDataParallelSE3(nn.Module):
         self._param_bucket = {}
         self._hooks = []
 
-        if self.world_size > 1:
+        if self.world_size > 1 and sync_params:
             self._broadcast_parameters()
         self._build_buckets(bucket_bytes)
         self._register_hooks()

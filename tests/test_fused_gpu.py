@@ -292,3 +292,24 @@ def test_fused_attention_one_headed_vs_eager():
     (out['0'].pow(2).mean() + out['1'].pow(2).mean()).backward()
     for d in gref:
         assert _rel_err(feats2[d].grad, gref[d]) < 1e-3, d
+
+
+@needs_gpu
+def test_knn_kernel_causal_vs_eager():
+    import os as _os
+    from se3_transformer_amd import SE3Transformer
+    torch.manual_seed(8)
+    model = SE3Transformer(dim=32, heads=2, dim_head=16, depth=1,
+                           num_degrees=2, num_neighbors=5, causal=True,
+                           attend_self=True, valid_radius=3.).to('cuda')
+    feats = torch.randn(1, 40, 32, device='cuda')
+    coors = torch.randn(1, 40, 3, device='cuda')
+    mask = torch.ones(1, 40, dtype=torch.bool, device='cuda')
+    _os.environ['SE3_EAGER_KNN'] = '1'
+    try:
+        ref = model(feats, coors, mask, return_type=0)
+    finally:
+        del _os.environ['SE3_EAGER_KNN']
+    out = model(feats, coors, mask, return_type=0)
+    err = (out - ref).abs().max().item()
+    assert err < 1e-4, f'causal knn mismatch: {err}'
