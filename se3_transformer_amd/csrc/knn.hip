@@ -39,7 +39,7 @@ knn_kernel(const float* __restrict__ coors,        // (b, n, 3)
     const float xi = coors[q * 3], yi = coors[q * 3 + 1], zi = coors[q * 3 + 2];
 
     int cnt = 0;  // entries in this lane's list
-    for (int s = 0; s < MAXK; ++s) ld[w][s][lane] = 3.0e38f;
+    for (int s = 0; s < MAXK; ++s) { ld[w][s][lane] = 3.0e38f; li[w][s][lane] = 0; }
     for (int j = lane; j < n; j += 64) {
         if (j == i) continue;
         if (causal && j >= i) continue;
@@ -72,7 +72,9 @@ knn_kernel(const float* __restrict__ coors,        // (b, n, 3)
             if (ov < v || (ov == v && ol < l)) { v = ov; l = ol; }
         }
         if (lane == l) {
-            sel[w][r] = li[w][head][lane];
+            // v == +inf means no candidate anywhere (e.g. causal row 0):
+            // emit node 0 with an invalid marker so the mask comes out 0.
+            sel[w][r] = (v < 3.0e38f) ? li[w][head][lane] : -1;
             ++head;
         }
     }
@@ -81,6 +83,8 @@ knn_kernel(const float* __restrict__ coors,        // (b, n, 3)
     // lanes 0..k-1: write the selected neighbors' geometry + validity
     if (lane < k) {
         int j = sel[w][lane];
+        bool valid = j >= 0;
+        if (!valid) j = (i == 0) ? (n > 1 ? 1 : 0) : 0;  // any real node != i
         float dx = xi - coors[((long)bi * n + j) * 3];
         float dy = yi - coors[((long)bi * n + j) * 3 + 1];
         float dz = zi - coors[((long)bi * n + j) * 3 + 2];
@@ -91,7 +95,7 @@ knn_kernel(const float* __restrict__ coors,        // (b, n, 3)
         out_rel[o * 3] = dx;
         out_rel[o * 3 + 1] = dy;
         out_rel[o * 3 + 2] = dz;
-        unsigned char ok = d <= radius;
+        unsigned char ok = valid && (d <= radius);
         if (nmask != nullptr)
             ok = ok & nmask[(long)bi * n + i] & nmask[(long)bi * n + j];
         out_m[o] = ok;
