@@ -105,3 +105,18 @@ def test_one_headed_kv_cuda_bf16():
     with torch.autocast(device_type='cuda', dtype=torch.bfloat16):
         out = model(feats, coors, mask, return_type=0)
     assert torch.isfinite(out).all()
+
+
+@needs_gpu
+def test_differentiable_coors_cuda():
+    """AF2-style refinement config: grads must flow to coordinates."""
+    model = SE3Transformer(dim=24, heads=2, dim_head=12, depth=1,
+                           num_degrees=2, output_degrees=2, num_neighbors=6,
+                           differentiable_coors=True).to('cuda')
+    feats = torch.randn(1, 40, 24, device='cuda')
+    coors = torch.randn(1, 40, 3, device='cuda', requires_grad=True)
+    mask = torch.ones(1, 40, dtype=torch.bool, device='cuda')
+    out = model(feats, coors, mask, return_type=1)
+    out.pow(2).mean().backward()
+    torch.cuda.synchronize()
+    assert coors.grad is not None and torch.isfinite(coors.grad).all()
