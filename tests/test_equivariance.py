@@ -186,3 +186,27 @@ def test_differentiable_coors_grads():
     out = model(feats, coors, mask, return_type=1)
     out.pow(2).sum().backward()
     assert coors.grad is not None and torch.isfinite(coors.grad).all()
+
+
+def test_pairwise_conv_reference_contract():
+    """PairwiseConv.forward (the reference's materialized per-edge kernel
+    matrix, se3_transformer_pytorch.py:326-343) must agree with the
+    streaming apply_fused path used by the framework. Reference layout:
+    kernel rows/cols are channel-major ((ch, m) flattening, :331,343) and
+    x is flattened the same way (:238)."""
+    import torch
+    from se3_transformer_amd.models.core import PairwiseConv
+    from se3_transformer_amd.utils import to_order
+
+    torch.manual_seed(0)
+    di, do, mi, mo, E = 1, 2, 8, 6, 50
+    pc = PairwiseConv(di, mi, do, mo, edge_dim=2)
+    F_, O, I = pc.num_freq, to_order(do), to_order(di)
+    ef = torch.randn(E, 3)
+    basis = {(di, do): torch.randn(E, O, I, F_)}
+    x = torch.randn(E, mi, I)
+
+    k = pc(ef, basis)                          # (E, mo*O, mi*I) channel-major
+    ref = torch.bmm(k, x.reshape(E, mi * I, 1)).view(E, mo, O)
+    out = pc.apply_fused(ef, basis, x)         # (E, mo, O)
+    assert (out - ref).abs().max() < 1e-4
