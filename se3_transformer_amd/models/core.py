@@ -286,6 +286,16 @@ class ConvSE3(nn.Module):
             self.self_interact = LinearSE3(fiber_in, fiber_out)
             self.self_interact_sum = ResidualSE3()
 
+    _stream_pool = {}
+
+    @staticmethod
+    def _streams(device, k=4):
+        pool = ConvSE3._stream_pool.get((device, k))
+        if pool is None:
+            pool = [torch.cuda.Stream(device=device) for _ in range(k)]
+            ConvSE3._stream_pool[(device, k)] = pool
+        return pool
+
     def forward(self, inp, edge_info, rel_dist=None, basis=None):
         neighbor_indices, neighbor_masks, edges = edge_info
         rel = rel_dist.unsqueeze(-1)
@@ -299,12 +309,36 @@ class ConvSE3(nn.Module):
             x = batched_index_select(inp[str(di)], neighbor_indices, dim=1)
             gathered[di] = x  # (b, n, k, mi, 2di+1)
 
+        pairs = list(self.fiber_in * self.fiber_out)
+        use_streams = (edge_feats.is_cuda and len(pairs) > 1
+                       and os.environ.get('SE3_STREAMS', '1') != '0')
+
+        pair_out = {}
+        if use_streams:
+            # the degree pairs are data-independent: fan them out over a small
+            # HIP stream pool so one pair's stalls overlap another's compute
+            cur = torch.cuda.current_stream()
+            pool = self._streams(edge_feats.device)
+            for idx, ((di, _mi), (do, _mo)) in enumerate(pairs):
+                s = pool[idx % len(pool)]
+                s.wait_stream(cur)
+                with torch.cuda.stream(s):
+                    pc = self.kernel_unary[f'({di},{do})']
+                    pair_out[(di, do)] = pc.apply_fused(edge_feats, basis, gathered[di])
+            for s in pool:
+                cur.wait_stream(s)
+            for t in pair_out.values():
+                t.record_stream(cur)
+        else:
+            for (di, _mi), (do, _mo) in pairs:
+                pc = self.kernel_unary[f'({di},{do})']
+                pair_out[(di, do)] = pc.apply_fused(edge_feats, basis, gathered[di])
+
         outputs = {}
         for do, _mo in self.fiber_out:
             acc = None
             for di, _mi in self.fiber_in:
-                pc = self.kernel_unary[f'({di},{do})']
-                out = pc.apply_fused(edge_feats, basis, gathered[di])
+                out = pair_out[(di, do)]
                 acc = out if acc is None else acc + out
 
             if self.pool:
