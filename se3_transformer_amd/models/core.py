@@ -310,8 +310,11 @@ class ConvSE3(nn.Module):
             gathered[di] = x  # (b, n, k, mi, 2di+1)
 
         pairs = list(self.fiber_in * self.fiber_out)
+        # opt-in: overlaps independent pairs on a stream pool, but is NOT
+        # compatible with hipGraph step capture (capture with forked streams
+        # hangs on ROCm 7.2) — default off
         use_streams = (edge_feats.is_cuda and len(pairs) > 1
-                       and os.environ.get('SE3_STREAMS', '1') != '0')
+                       and os.environ.get('SE3_STREAMS') == '1')
 
         pair_out = {}
         if use_streams:
