@@ -1,0 +1,45 @@
+"""Guard the driver contract: `python bench.py --gpus N --steps K --warmup W`
+must print exactly one JSON line with the required fields, single- and
+multi-process (gloo on CPU)."""
+import json
+import os
+import subprocess
+import sys
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+TINY = ['--steps', '1', '--warmup', '0', '--points', '32', '--dim', '32',
+        '--heads', '2', '--dim-head', '16', '--depth', '1',
+        '--num-degrees', '2', '--num-neighbors', '4', '--dtype', 'fp32']
+
+REQUIRED = {'metric', 'value', 'unit', 'n_gpus', 'steps', 'warmup',
+            'ms_per_step', 'higher_is_better', 'scaling', 'vs_baseline',
+            'dtype', 'data', 'config'}
+
+
+def _check_line(out):
+    lines = [l for l in out.strip().splitlines() if l.startswith('{')]
+    assert len(lines) == 1, f'expected exactly one JSON line, got: {out!r}'
+    d = json.loads(lines[0])
+    assert REQUIRED <= set(d), REQUIRED - set(d)
+    assert d['value'] > 0 and d['ms_per_step'] > 0
+    assert d['config']['global_batch'] == d['n_gpus'] * 1
+    return d
+
+
+def test_bench_single_process():
+    r = subprocess.run([sys.executable, 'bench.py', '--gpus', '1'] + TINY,
+                       capture_output=True, text=True, cwd=REPO, timeout=600)
+    assert r.returncode == 0, r.stderr[-2000:]
+    d = _check_line(r.stdout)
+    assert d['n_gpus'] == 1
+
+
+def test_bench_two_process_gloo():
+    r = subprocess.run(
+        [sys.executable, '-m', 'torch.distributed.run', '--nnodes=1',
+         '--nproc-per-node', '2', '--master-addr', '127.0.0.1',
+         '--master-port', '29719', 'bench.py', '--gpus', '2'] + TINY,
+        capture_output=True, text=True, cwd=REPO, timeout=900)
+    assert r.returncode == 0, r.stderr[-2000:]
+    d = _check_line(r.stdout)
+    assert d['n_gpus'] == 2 and d['config']['parallelism'] == 'dp2'
