@@ -9,7 +9,7 @@ from __future__ import annotations
 
 import contextlib
 import time
-from functools import wraps
+from functools import lru_cache, wraps
 
 import torch
 
@@ -173,3 +173,60 @@ def fast_split(arr, splits, dim=0):
         size = base + (1 if i < rem else 0)
         yield torch.narrow(arr, dim, start, size)
         start += size
+
+
+def cache(cache_dict, key_fn):
+    """In-RAM memoization with a custom key function (reference
+    utils.py:135-147)."""
+    def cache_inner(fn):
+        @wraps(fn)
+        def inner(*args, **kwargs):
+            key = key_fn(*args, **kwargs)
+            if key in cache_dict:
+                return cache_dict[key]
+            res = fn(*args, **kwargs)
+            cache_dict[key] = res
+            return res
+        return inner
+    return cache_inner
+
+
+def cache_dir(dirname, maxsize=128):
+    """On-disk cache decorator (reference utils.py:151-206). Keeps the
+    reference contract (results keyed by repr of the call, persisted under
+    `dirname`, lru_cache front, CLEAR_CACHE env disables) but uses atomic
+    tmp+rename writes instead of a FileLock."""
+    import hashlib
+    import os
+    import tempfile
+
+    import torch as _torch
+
+    def decorator(func):
+        if os.environ.get('CLEAR_CACHE') is not None:
+            return func
+        root = os.path.expanduser(dirname)
+
+        @lru_cache(maxsize=maxsize)
+        @wraps(func)
+        def wrapper(*args, **kwargs):
+            key = repr((args, tuple(sorted(kwargs.items()))))
+            h = hashlib.sha1(key.encode()).hexdigest()[:24]
+            path = os.path.join(root, f'{func.__name__}.{h}.pt')
+            if os.path.exists(path):
+                try:
+                    return _torch.load(path, weights_only=False)
+                except Exception:
+                    pass
+            result = func(*args, **kwargs)
+            try:
+                os.makedirs(root, exist_ok=True)
+                fd, tmp = tempfile.mkstemp(dir=root)
+                os.close(fd)
+                _torch.save(result, tmp)
+                os.replace(tmp, path)
+            except OSError:
+                pass
+            return result
+        return wrapper
+    return decorator
