@@ -132,9 +132,11 @@ def main():
 
     # hipGraph capture: the step is shape-static, so capture once and replay —
     # removes the Python/launch-gap overhead between the ~10k kernels per step.
-    # For world > 1 the RCCL all-reduces are captured too (stream-ordered);
-    # if capture fails for any reason we fall back to the eager step.
-    graph_mode = (args.graph if args.graph is not None else True) and use_cuda
+    # Default on for single-GPU; for world > 1 pass --graph explicitly (RCCL
+    # collectives are stream-captured, but multi-rank capture is not validated
+    # on this RCCL build, and a capture hang would kill a scaling run).
+    graph_mode = (args.graph if args.graph is not None else world == 1) \
+        and use_cuda
     if graph_mode:
         try:
             for _ in range(2):      # allocator warmup before capture
