@@ -25,18 +25,40 @@ from se3_transformer_amd import SE3Transformer
 from se3_transformer_amd.parallel import DistributedDataParallelSE3, setup_distributed
 
 
+# BASELINE.json named configs beyond the headline (selected with --preset):
+#   qm9        — 29-atom molecules, discrete edges + sparse-adjacency attention
+#   af2        — 256-residue refinement, output_degrees=2, differentiable_coors
+#   rev2048    — reversible trunk, 2048-node point cloud
+PRESETS = {
+    'qm9': dict(points=29, dim=128, heads=4, dim_head=32, depth=4,
+                num_degrees=3, num_neighbors=8, batch=16,
+                model_kwargs=dict(num_edge_tokens=4, edge_dim=16,
+                                  attend_sparse_neighbors=True,
+                                  num_adj_degrees=2, adj_dim=4,
+                                  num_neighbors=0)),
+    'af2': dict(points=256, dim=256, heads=8, dim_head=32, depth=4,
+                num_degrees=2, num_neighbors=12, batch=1,
+                model_kwargs=dict(output_degrees=2, differentiable_coors=True)),
+    'rev2048': dict(points=2048, dim=256, heads=8, dim_head=32, depth=6,
+                    num_degrees=3, num_neighbors=8, batch=1,
+                    model_kwargs=dict(reversible=True)),
+}
+
+
 def build_model(args, device):
+    kwargs = dict(
+        dim=args.dim,
+        heads=args.heads,
+        dim_head=args.dim_head,
+        depth=args.depth,
+        num_degrees=args.num_degrees,
+        valid_radius=args.valid_radius,
+        num_neighbors=args.num_neighbors,
+        attend_self=True,
+    )
+    kwargs.update(getattr(args, 'model_kwargs', {}))
     with torch.device(device):
-        model = SE3Transformer(
-            dim=args.dim,
-            heads=args.heads,
-            dim_head=args.dim_head,
-            depth=args.depth,
-            num_degrees=args.num_degrees,
-            valid_radius=args.valid_radius,
-            num_neighbors=args.num_neighbors,
-            attend_self=True,
-        )
+        model = SE3Transformer(**kwargs)
     return model
 
 
@@ -62,7 +84,13 @@ def main():
     p.add_argument('--device', type=str, default=None)
     p.add_argument('--timers', action='store_true',
                    help='report per-phase (fwd/bwd/opt) hipEvent timings')
+    p.add_argument('--preset', default=None, choices=sorted(PRESETS),
+                   help='one of the BASELINE.json named configs')
     args = p.parse_args()
+    args.model_kwargs = {}
+    if args.preset:
+        for k, v in PRESETS[args.preset].items():
+            setattr(args, k, v)
 
     rank, world, local_rank = setup_distributed()
     use_cuda = torch.cuda.is_available()
@@ -89,6 +117,12 @@ def main():
     coors = (torch.randn(args.batch, args.points, 3, generator=g) * 2.0).to(device)
     mask = torch.ones(args.batch, args.points, dtype=torch.bool, device=device)
     target = torch.randn(args.batch, args.points, args.dim, generator=g).to(device)
+    fwd_extra = {}
+    if args.preset == 'qm9':
+        i = torch.arange(args.points, device=device)
+        fwd_extra['adj_mat'] = (i[:, None] - i[None, :]).abs() == 1  # molecular chain
+        fwd_extra['edges'] = torch.randint(0, 4, (args.batch, args.points, args.points),
+                                           generator=g, device='cpu').to(device)
 
     autocast_dtype = torch.bfloat16 if args.dtype == 'bf16' else torch.float32
     autocast_enabled = args.dtype == 'bf16'
@@ -106,7 +140,7 @@ def main():
             phase_events[0].record()
         with torch.autocast(device_type=device.type, dtype=autocast_dtype,
                             enabled=autocast_enabled):
-            out = runner(feats, coors, mask, return_type=0)
+            out = runner(feats, coors, mask, return_type=0, **fwd_extra)
             loss = (out.float() - target).pow(2).mean()
         if phase_events:
             phase_events[1].record()
@@ -188,8 +222,11 @@ def main():
     samples_per_sec = (args.batch * world * args.steps) / elapsed
 
     if rank == 0:
+        metric = 'train samples/sec (fwd+bwd), 1024-pt dim=512 depth=6 num_degrees=4'
+        if args.preset:
+            metric = f'train samples/sec (fwd+bwd), preset={args.preset}'
         print(json.dumps({
-            'metric': 'train samples/sec (fwd+bwd), 1024-pt dim=512 depth=6 num_degrees=4',
+            'metric': metric,
             'value': samples_per_sec,
             'unit': 'samples/sec',
             'n_gpus': world,
@@ -210,6 +247,7 @@ def main():
                 'parallelism': f'dp{world}',
                 'n_params': n_params,
                 'optimizer': 'sgd',
+                'preset': args.preset,
             },
         }))
 
