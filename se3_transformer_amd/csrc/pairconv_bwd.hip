@@ -197,36 +197,67 @@ pairconv_bwd_dw_kernel(const __bf16* __restrict__ Gt,  // (mo, O, E)
 #pragma unroll
         for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
+    // T14 register-staged pipeline: chunk ec+1's global loads are issued
+    // right after chunk ec's LDS image is written, so HBM/L2 latency hides
+    // under the dR + MFMA phase instead of serializing each chunk.
+    constexpr int UU = (32 * O * 32 / 8 + NT - 1) / NT;   // u 16B units/thread
+    constexpr int GU = 1;                                  // g units/thread
+    bf16x8 u_reg[UU], g_reg[GU], h_reg;
     const int nec = (E + 31) / 32;
-    for (int ec = 0; ec < nec; ++ec) {
+
+    auto load_chunk = [&](int ec) {
         const int e0 = ec * 32;
-        __syncthreads();
-        // stage u chunk [32c][O][32e] and g tile [4m][O][32e] (16B units)
-        for (int i = tid; i < (32 * O * 32) / 8; i += NT) {
-            int ro = i >> 2, eu = (i & 3) * 8;
-            const __bf16* src = Ut + ((size_t)(cb * 32 + ro / O) * O + (ro % O)) * E + e0 + eu;
-            bf16x8 v;
-            if (e0 + eu + 8 <= E) v = *reinterpret_cast<const bf16x8*>(src);
-            else { for (int j = 0; j < 8; ++j) v[j] = (e0 + eu + j < E) ? src[j] : (__bf16)0.f; }
-            *reinterpret_cast<bf16x8*>(u_lds + (size_t)ro * 32 + eu) = v;
+#pragma unroll
+        for (int t = 0; t < UU; ++t) {
+            int i = tid + t * NT;
+            u_reg[t] = bf16x8(0);
+            if (i < (32 * O * 32) / 8) {
+                int ro = i >> 2, eu = (i & 3) * 8;
+                const __bf16* src = Ut + ((size_t)(cb * 32 + ro / O) * O + (ro % O)) * E + e0 + eu;
+                if (e0 + eu + 8 <= E) u_reg[t] = *reinterpret_cast<const bf16x8*>(src);
+                else { for (int j = 0; j < 8; ++j) u_reg[t][j] = (e0 + eu + j < E) ? src[j] : (__bf16)0.f; }
+            }
         }
-        for (int i = tid; i < (4 * O * 32) / 8; i += NT) {
-            int ro = i >> 2, eu = (i & 3) * 8;
-            const __bf16* src = Gt + ((size_t)(mb * 4 + ro / O) * O + (ro % O)) * E + e0 + eu;
-            bf16x8 v;
-            if (e0 + eu + 8 <= E) v = *reinterpret_cast<const bf16x8*>(src);
-            else { for (int j = 0; j < 8; ++j) v[j] = (e0 + eu + j < E) ? src[j] : (__bf16)0.f; }
-            *reinterpret_cast<bf16x8*>(g_lds + (size_t)ro * 32 + eu) = v;
+        {
+            int i = tid;
+            g_reg[0] = bf16x8(0);
+            if (i < (4 * O * 32) / 8) {
+                int ro = i >> 2, eu = (i & 3) * 8;
+                const __bf16* src = Gt + ((size_t)(mb * 4 + ro / O) * O + (ro % O)) * E + e0 + eu;
+                if (e0 + eu + 8 <= E) g_reg[0] = *reinterpret_cast<const bf16x8*>(src);
+                else { for (int j = 0; j < 8; ++j) g_reg[0][j] = (e0 + eu + j < E) ? src[j] : (__bf16)0.f; }
+            }
         }
-        // stage H^T chunk [128k][32e]
-        for (int i = tid; i < (128 * 32) / 8; i += NT) {
-            int k = i >> 2, eu = (i & 3) * 8;
+        {
+            int k = tid >> 2, eu = (tid & 3) * 8;
             const __bf16* src = Ht + (size_t)k * E + e0 + eu;
-            bf16x8 v;
-            if (e0 + eu + 8 <= E) v = *reinterpret_cast<const bf16x8*>(src);
-            else { for (int j = 0; j < 8; ++j) v[j] = (e0 + eu + j < E) ? src[j] : (__bf16)0.f; }
-            *reinterpret_cast<bf16x8*>(h_lds + (size_t)k * 32 + eu) = v;
+            h_reg = bf16x8(0);
+            if (e0 + eu + 8 <= E) h_reg = *reinterpret_cast<const bf16x8*>(src);
+            else { for (int j = 0; j < 8; ++j) h_reg[j] = (e0 + eu + j < E) ? src[j] : (__bf16)0.f; }
         }
+    };
+
+    load_chunk(0);
+    for (int ec = 0; ec < nec; ++ec) {
+        __syncthreads();   // previous MFMA done reading the LDS images
+        // write the staged registers for chunk ec
+#pragma unroll
+        for (int t = 0; t < UU; ++t) {
+            int i = tid + t * NT;
+            if (i < (32 * O * 32) / 8) {
+                int ro = i >> 2, eu = (i & 3) * 8;
+                *reinterpret_cast<bf16x8*>(u_lds + (size_t)ro * 32 + eu) = u_reg[t];
+            }
+        }
+        if (tid < (4 * O * 32) / 8) {
+            int ro = tid >> 2, eu = (tid & 3) * 8;
+            *reinterpret_cast<bf16x8*>(g_lds + (size_t)ro * 32 + eu) = g_reg[0];
+        }
+        {
+            int k = tid >> 2, eu = (tid & 3) * 8;
+            *reinterpret_cast<bf16x8*>(h_lds + (size_t)k * 32 + eu) = h_reg;
+        }
+        if (ec + 1 < nec) load_chunk(ec + 1);   // issue next loads early
         __syncthreads();
         // cooperative dR^T tile [128n][32e] (e-pairs, packed)
         for (int i = tid; i < (128 * 32) / 2; i += NT) {
@@ -246,17 +277,14 @@ pairconv_bwd_dw_kernel(const __bf16* __restrict__ Gt,  // (mo, O, E)
         __syncthreads();
         // MFMA: dW_tile += dR^T(128n x 32e) @ H(32e x 128k)
 #pragma unroll
-        for (int es = 0; es < 1; ++es) {
+        for (int nf = 0; nf < 2; ++nf) {
+            bf16x8 a = *reinterpret_cast<const bf16x8*>(
+                dr_lds + (size_t)(wn * 32 + nf * 16 + l15) * 32 + l4 * 8);
 #pragma unroll
-            for (int nf = 0; nf < 2; ++nf) {
-                bf16x8 a = *reinterpret_cast<const bf16x8*>(
-                    dr_lds + (size_t)(wn * 32 + nf * 16 + l15) * 32 + l4 * 8);
-#pragma unroll
-                for (int kf = 0; kf < 4; ++kf) {
-                    bf16x8 b = *reinterpret_cast<const bf16x8*>(
-                        h_lds + (size_t)(wk * 64 + kf * 16 + l15) * 32 + l4 * 8);
-                    acc[nf][kf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nf][kf], 0, 0, 0);
-                }
+            for (int kf = 0; kf < 4; ++kf) {
+                bf16x8 b = *reinterpret_cast<const bf16x8*>(
+                    h_lds + (size_t)(wk * 64 + kf * 16 + l15) * 32 + l4 * 8);
+                acc[nf][kf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[nf][kf], 0, 0, 0);
             }
         }
     }
