@@ -210,3 +210,27 @@ def test_pairwise_conv_reference_contract():
     ref = torch.bmm(k, x.reshape(E, mi * I, 1)).view(E, mo, O)
     out = pc.apply_fused(ef, basis, x)         # (E, mo, O)
     assert (out - ref).abs().max() < 1e-4
+
+
+def test_equivariance_num_degrees_4_f64():
+    """Full headline degree range (0..3; SH up to J=6, all 16 degree pairs)
+    in float64, mirroring the reference's strictest tolerance regime
+    (tests/test_equivariance.py:234-260 uses f64 for the hardest case)."""
+    import torch
+    from se3_transformer_amd import SE3Transformer
+    from se3_transformer_amd.ops.wigner import rot
+    from se3_transformer_amd.utils import torch_default_dtype
+
+    with torch_default_dtype(torch.float64):
+        torch.manual_seed(0)
+        model = SE3Transformer(dim=8, heads=2, dim_head=4, depth=1,
+                               attend_self=True, num_neighbors=4,
+                               num_degrees=4, output_degrees=2)
+        feats = torch.randn(1, 12, 8)
+        coors = torch.randn(1, 12, 3)
+        mask = torch.ones(1, 12, dtype=torch.bool)
+        R = rot(23., 117., 195.).to(torch.float64)
+        out1 = model(feats, coors @ R, mask, return_type=1)
+        out2 = model(feats, coors, mask, return_type=1) @ R
+        diff = (out1 - out2).abs().max()
+        assert diff < 1e-8, f'degree-4 equivariance violated: {diff}'
