@@ -173,3 +173,41 @@ def test_basis_kernel_equivariance():
     d_i = wigner_d_from_matrix(d_in, m_perm)
     rhs = torch.einsum('oi,nij,pj->nop', d_o, k0, d_i)
     assert (k1 - rhs).abs().max() < 1e-8
+
+
+def test_utils_fast_split_and_fourier():
+    """Reference utils parity: fast_split chunking (utils.py:85-94) and
+    fourier_encode (utils.py:96-104)."""
+    import torch
+    from se3_transformer_amd.utils import fast_split, fourier_encode
+
+    t = torch.arange(10).float().unsqueeze(-1)
+    chunks = list(fast_split(t, 4, dim=0))
+    assert sum(c.shape[0] for c in chunks) == 10
+    assert torch.equal(torch.cat(chunks, dim=0), t)
+
+    x = torch.randn(3, 5)
+    enc = fourier_encode(x, num_encodings=4, include_self=True)
+    assert enc.shape == (3, 5 * (2 * 4 + 1))
+    enc2 = fourier_encode(x, num_encodings=2, include_self=False, flatten=False)
+    assert enc2.shape == (3, 5, 4)
+
+
+def test_utils_batched_index_select_masked_mean():
+    import torch
+    from se3_transformer_amd.utils import batched_index_select, masked_mean
+
+    v = torch.randn(2, 6, 3, 5)
+    idx = torch.randint(0, 6, (2, 4))
+    out = batched_index_select(v, idx, dim=1)
+    for b in range(2):
+        for i in range(4):
+            assert torch.equal(out[b, i], v[b, idx[b, i]])
+
+    t = torch.randn(2, 4, 3)
+    m = torch.tensor([[True, True, False, False], [True, False, False, False]])
+    mm = masked_mean(t, m, dim=1)
+    assert torch.allclose(mm[0], t[0, :2].mean(dim=0), atol=1e-6)
+    assert torch.allclose(mm[1], t[1, :1].mean(dim=0), atol=1e-6)
+    # functional: input not mutated (reference masked_fill_ mutates — SURVEY §2.4)
+    assert torch.isfinite(t).all()
