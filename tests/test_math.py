@@ -186,10 +186,12 @@ def test_utils_fast_split_and_fourier():
     assert sum(c.shape[0] for c in chunks) == 10
     assert torch.equal(torch.cat(chunks, dim=0), t)
 
-    x = torch.randn(3, 5)
+    # reference flatten keeps the first three dims: 'b m n ... -> b m n (...)'
+    x = torch.randn(3, 5, 2, 4)
     enc = fourier_encode(x, num_encodings=4, include_self=True)
-    assert enc.shape == (3, 5 * (2 * 4 + 1))
-    enc2 = fourier_encode(x, num_encodings=2, include_self=False, flatten=False)
+    assert enc.shape == (3, 5, 2, 4 * (2 * 4 + 1))
+    enc2 = fourier_encode(torch.randn(3, 5), num_encodings=2,
+                          include_self=False, flatten=False)
     assert enc2.shape == (3, 5, 4)
 
 
