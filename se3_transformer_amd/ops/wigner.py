@@ -29,6 +29,7 @@ from ..utils import cast_torch_tensor, to_order, torch_default_dtype
 from .sh import sh_packed_from_angles, sh_offset
 
 __all__ = [
+    'wigner_d_matrix', 'z_rot_mat',
     'rot_z', 'rot_y', 'rot', 'x_to_alpha_beta', 'compose',
     'spherical_harmonics', 'irr_repr', 'wigner_d',
 ]
@@ -144,6 +145,29 @@ def wigner_d_from_matrix(order: int, m3: torch.Tensor) -> torch.Tensor:
 def wigner_d(order: int, alpha, beta, gamma) -> torch.Tensor:
     """Real Wigner-D matrix D_order(alpha, beta, gamma), float64, [2l+1, 2l+1]."""
     return _wigner_d_cached(order, _angles_key(alpha), _angles_key(beta), _angles_key(gamma))
+
+
+def z_rot_mat(angle, l: int) -> torch.Tensor:
+    """Rotation about z in the order-l irrep basis (reference irr_repr.py:32-42)."""
+    angle = torch.as_tensor(angle, dtype=torch.get_default_dtype())
+    n = 2 * l + 1
+    inds = torch.arange(n)
+    reversed_inds = torch.arange(2 * l, -1, -1)
+    frequencies = torch.arange(l, -l - 1, -1, dtype=angle.dtype)
+    m = torch.zeros(n, n, dtype=angle.dtype)
+    m[inds, reversed_inds] = torch.sin(frequencies * angle)
+    m[inds, inds] = torch.cos(frequencies * angle)
+    return m
+
+
+def wigner_d_matrix(degree: int, alpha, beta, gamma, dtype=None, device=None):
+    """Reference-named alias of wigner_d (irr_repr.py:22-30)."""
+    d = wigner_d(degree, alpha, beta, gamma)
+    if dtype is not None:
+        d = d.to(dtype)
+    if device is not None:
+        d = d.to(device)
+    return d
 
 
 def irr_repr(order: int, alpha, beta, gamma, dtype=None) -> torch.Tensor:
