@@ -234,3 +234,32 @@ def test_equivariance_num_degrees_4_f64():
         out2 = model(feats, coors, mask, return_type=1) @ R
         diff = (out1 - out2).abs().max()
         assert diff < 1e-8, f'degree-4 equivariance violated: {diff}'
+
+
+def test_neighbor_mask_argument():
+    """User-supplied neighbor_mask restricts selection (reference
+    se3_transformer_pytorch.py:1250-1257)."""
+    import torch
+    from se3_transformer_amd import SE3Transformer
+    torch.manual_seed(0)
+    model = SE3Transformer(dim=16, depth=1, num_degrees=2, num_neighbors=3,
+                           heads=2, dim_head=8, attend_self=True)
+    b, n = 1, 10
+    feats = torch.randn(b, n, 16)
+    coors = torch.randn(b, n, 3)
+    mask = torch.ones(b, n, dtype=torch.bool)
+    nmask = torch.rand(b, n, n) > 0.3
+    out = model(feats, coors, mask, neighbor_mask=nmask, return_type=0)
+    assert out.shape == (b, n, 16) and torch.isfinite(out).all()
+
+
+def test_return_pooled():
+    import torch
+    from se3_transformer_amd import SE3Transformer
+    model = SE3Transformer(dim=16, depth=1, num_degrees=2, num_neighbors=3,
+                           heads=2, dim_head=8)
+    feats = torch.randn(2, 8, 16)
+    coors = torch.randn(2, 8, 3)
+    mask = torch.ones(2, 8, dtype=torch.bool)
+    out = model(feats, coors, mask, return_pooled=True, return_type=0)
+    assert out.shape == (2, 16)
