@@ -43,3 +43,14 @@ def test_bench_two_process_gloo():
     assert r.returncode == 0, r.stderr[-2000:]
     d = _check_line(r.stdout)
     assert d['n_gpus'] == 2 and d['config']['parallelism'] == 'dp2'
+
+
+def test_denoise_example_runs():
+    """The training example must run one accumulation cycle on CPU."""
+    import subprocess
+    import sys
+    r = subprocess.run([sys.executable, 'examples/denoise.py', '--steps', '1',
+                        '--length', '16'],
+                       capture_output=True, text=True, cwd=REPO, timeout=600)
+    assert r.returncode == 0, r.stderr[-1500:]
+    assert 'loss:' in r.stdout
