@@ -85,10 +85,16 @@ knn_kernel(const float* __restrict__ coors,        // (b, n, 3)
         int j = sel[w][lane];
         bool valid = j >= 0;
         if (!valid) j = (i == 0) ? (n > 1 ? 1 : 0) : 0;  // any real node != i
-        float dx = xi - coors[((long)bi * n + j) * 3];
-        float dy = yi - coors[((long)bi * n + j) * 3 + 1];
-        float dz = zi - coors[((long)bi * n + j) * 3 + 2];
-        float d = sqrtf(dx * dx + dy * dy + dz * dz);
+        float dx = 0.f, dy = 0.f, dz = 0.f, d = 0.f;
+        if (valid) {
+            // invalid slots (causal row 0, k > candidates) get index j with
+            // ZERO geometry so consumers that ignore the mask see no
+            // fabricated edge; the output mask is 0 either way.
+            dx = xi - coors[((long)bi * n + j) * 3];
+            dy = yi - coors[((long)bi * n + j) * 3 + 1];
+            dz = zi - coors[((long)bi * n + j) * 3 + 2];
+            d = sqrtf(dx * dx + dy * dy + dz * dz);
+        }
         long o = q * k + lane;
         out_idx[o] = j;
         out_dist[o] = d;

@@ -314,7 +314,10 @@ class ConvSE3(nn.Module):
         # compatible with hipGraph step capture (capture with forked streams
         # hangs on ROCm 7.2) — default off
         use_streams = (edge_feats.is_cuda and len(pairs) > 1
-                       and os.environ.get('SE3_STREAMS') == '1')
+                       and os.environ.get('SE3_STREAMS') == '1'
+                       # forked streams hang under hipGraph capture on ROCm
+                       # 7.2: fall back to the single-stream loop mid-capture
+                       and not torch.cuda.is_current_stream_capturing())
 
         pair_out = {}
         if use_streams:

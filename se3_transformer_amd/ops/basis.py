@@ -59,9 +59,8 @@ def _null_space_1d(mats, eps=1e-9) -> torch.Tensor:
     """The single common null vector of the stacked matrices (f64 SVD)."""
     a = torch.cat(mats, dim=0)
     _, s, vh = torch.linalg.svd(a)
-    null = vh[s.shape[0] - 1:] if s[-1] < eps else vh[0:0]
     # rows of vh beyond rank are the null space; with a 1-D null space the
-    # smallest singular value row is the solution
+    # smallest-singular-value row is the solution
     assert s[-1] < eps and (s.shape[0] < 2 or s[-2] > eps), \
         f'intertwiner null space is not 1-dimensional (singular values tail: {s[-3:]})'
     return vh[-1]
@@ -145,10 +144,19 @@ def _qj_transposed(J: int, d_in: int, d_out: int) -> torch.Tensor:
 _qj_dev_cache = {}
 
 
+def _canon_device(device) -> str:
+    """Canonical device key: 'cuda' and 'cuda:0' must not duplicate cache
+    entries."""
+    d = torch.device(device)
+    if d.type == 'cuda' and d.index is None:
+        d = torch.device('cuda', torch.cuda.current_device())
+    return str(d)
+
+
 def _qj_t_on(J: int, d_in: int, d_out: int, device, dtype) -> torch.Tensor:
     """Device/dtype-resident Q_J^T (cached so steady-state forwards — and
     hipGraph capture — never issue host-to-device copies)."""
-    key = (J, d_in, d_out, device, dtype)
+    key = (J, d_in, d_out, _canon_device(device), dtype)
     t = _qj_dev_cache.get(key)
     if t is None:
         t = _qj_transposed(J, d_in, d_out).to(device=device, dtype=dtype)
@@ -196,7 +204,7 @@ _sh_tables_cache = {}
 def _sh_basis_tables(max_degree: int, device):
     """Device-resident Q_J^T concat + per-pair meta + SH norm table for the
     fused sh_basis HIP kernel (cached; hipGraph-capture safe)."""
-    key = (max_degree, device)
+    key = (max_degree, _canon_device(device))
     hit = _sh_tables_cache.get(key)
     if hit is not None:
         return hit

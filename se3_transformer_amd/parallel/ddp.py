@@ -69,8 +69,11 @@ class DistributedDataParallelSE3(nn.Module):
                  process_group=None, average: bool = True,
                  grad_compression: str = 'none', sync_params: bool = True):
         """grad_compression='bf16' all-reduces a bf16 copy of each bucket
-        (halves xGMI traffic; fp32 master grads are restored from the reduced
-        bf16 values). sync_params=False skips the initial parameter broadcast
+        (halves xGMI traffic; fp32 master grads are then OVERWRITTEN by the
+        reduced bf16 values — a one-shot rounding with relative error
+        <= 2^-8 per element and no error feedback across steps, acceptable
+        for SGD/Adam at bf16-compute scale but NOT bit-equal to fp32
+        reduction). sync_params=False skips the initial parameter broadcast
         (for ranks that already hold identical weights, e.g. seeded init)."""
         super().__init__()
         assert grad_compression in ('none', 'bf16')

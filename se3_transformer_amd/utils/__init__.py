@@ -215,7 +215,9 @@ def cache_dir(dirname, maxsize=128):
             path = os.path.join(root, f'{func.__name__}.{h}.pt')
             if os.path.exists(path):
                 try:
-                    return _torch.load(path, weights_only=False)
+                    # cached values are tensors/tuples of tensors; refuse
+                    # arbitrary pickles from the user-writable cache dir
+                    return _torch.load(path, weights_only=True)
                 except Exception:
                     pass
             result = func(*args, **kwargs)
