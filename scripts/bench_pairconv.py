@@ -54,6 +54,26 @@ def main():
 
     from se3_transformer_amd.ops.fused import _pack_w_dh, _pack_w_fwd
     P = _pack_w_fwd(W, mo, miF)
+
+    if hasattr(_C, 'pack_w_both'):
+        Pf2 = torch.empty(N * K, dtype=torch.bfloat16, device=dev)
+        Pdh2 = torch.empty(N * K, dtype=torch.bfloat16, device=dev)
+        _C.pack_w_both(W, Pf2, Pdh2, mo)
+        okf = torch.equal(Pf2.view_as(P), P)
+        okd = torch.equal(Pdh2.view(_pack_w_dh(W, mo, miF).shape),
+                          _pack_w_dh(W, mo, miF))
+        ms = timeit(lambda: _C.pack_w_both(W, Pf2, Pdh2, mo))
+        gb = (N * K * 2 * 3) / 1e9   # read W bf16 + write both
+        print(f'pack_w_both: {ms:8.3f} ms  {gb/ms:7.2f} TB/s  '
+              f'parity fwd={okf} dh={okd}')
+
+    if os.environ.get('SE3_SWEEP_UU'):
+        for uu in ('0', '1', '2', '4'):
+            os.environ['SE3_FWD_UU'] = uu
+            ms = timeit(lambda: _C.pairconv_fwd(H, P, Ut, out, mo))
+            print(f'fwd UU={uu} ({di},{do}): {ms:8.3f} ms  '
+                  f'{gemm_fl/ms/1e9:7.1f} TF/s (gemm)')
+        del os.environ['SE3_FWD_UU']
     if args.only:
         fn = {'fwd': lambda: _C.pairconv_fwd(H, P, Ut, out, mo),
               'dh': lambda: _C.pairconv_bwd_dh(gt, Ut, _pack_w_dh(W, mo, miF),

@@ -21,9 +21,16 @@
 //
 // Tile: block = 512 threads (8 waves) owns (64 edges) x (8 mo); loops over
 // miF in chunks of 32; per chunk an MFMA GEMM of (256 n-rows x 64 e-cols, K=128)
-// with n = 8mo x 32urow, then a VALU contraction against u_lds into an LDS
-// partial accumulator. Each (e, mo) output is owned by exactly one block:
+// with n = 8mo x 32urow, then a VALU contraction against u_lds into
+// per-lane register accumulators (the cross-lane l4 reduce is deferred to
+// after the chunk loop). Each (e, mo) output is owned by exactly one block:
 // no atomics anywhere.
+//
+// Software pipeline (round 2): the next chunk's u tile is loaded into
+// registers while the current chunk's MFMA + epilogue run, so the
+// u-stage's HBM/L2 latency hides under compute instead of serializing
+// each chunk (the round-1 PMC analysis measured ~3.5k-cycle wave stalls
+// per chunk from the serialized u-stage -> W-frags round trips).
 
 #include <torch/extension.h>
 #include <hip/hip_runtime.h>
@@ -44,7 +51,7 @@ __device__ __forceinline__ float bf16_to_f32(unsigned short u) {
     return v.f;
 }
 
-template <int O>
+template <int O, int UU>   // UU = 16B u-units register-staged per thread
 __global__ void __launch_bounds__(NTHREADS, 4)   // cap VGPR<=128: 2 blocks/CU
 pairconv_fwd_kernel(const __bf16* __restrict__ H,
                     const __bf16* __restrict__ P,   // packed W: [mo/8][miF/32][wm4][mf4][kit4][lane64][8]
@@ -96,26 +103,60 @@ pairconv_fwd_kernel(const __bf16* __restrict__ H,
         // zero partial accumulator
         for (int i = tid; i < BLK_E * BLK_MO * O; i += NTHREADS) part[i] = 0.f;
     }
+
+    constexpr int UTOT = (UCHUNK * O * BLK_E) / 8;   // u chunk in 16B units
+    bf16x8 u_reg[UU > 0 ? UU : 1];
+    auto load_u = [&](int c) {
+        const int uc0 = c * UCHUNK;
+#pragma unroll
+        for (int t = 0; t < UU; ++t) {
+            int i = tid + t * NTHREADS;
+            if (i < UTOT) {
+                int ro = i >> 3;            // (urow*O + o)
+                int eu = (i & 7) * 8;       // e offset within 64
+                const __bf16* src = Ut + ((size_t)(uc0 + (ro / O)) * O + (ro % O)) * E + e0 + eu;
+                if (e0 + eu + 8 <= E) {
+                    u_reg[t] = *reinterpret_cast<const bf16x8*>(src);
+                } else {
+                    bf16x8 v(0);
+                    for (int j = 0; j < 8; ++j)
+                        if (e0 + eu + j < E) v[j] = src[j];
+                    u_reg[t] = v;
+                }
+            }
+        }
+    };
+
+    load_u(0);
     __syncthreads();
 
     const int nchunks = miF / UCHUNK;
     for (int c = 0; c < nchunks; ++c) {
-        const int uc0 = c * UCHUNK;
-
-        // ---- stage u chunk: u_lds[urow][o][e] <- Ut[(uc0+urow)*O + o][e0..e0+64]
-        for (int i = tid; i < (UCHUNK * O * BLK_E) / 8; i += NTHREADS) { // 16B units
-            int ro = i >> 3;            // (urow*O + o)
-            int eu = (i & 7) * 8;       // e offset within 64
-            const __bf16* src = Ut + ((size_t)(uc0 + (ro / O)) * O + (ro % O)) * E + e0 + eu;
+        // ---- commit the staged u chunk: u_lds[urow][o][e]
+#pragma unroll
+        for (int t = 0; t < UU; ++t) {
+            int i = tid + t * NTHREADS;
+            if (i < UTOT) {
+                int ro = i >> 3, eu = (i & 7) * 8;
+                *reinterpret_cast<bf16x8*>(u_lds + (size_t)ro * BLK_E + eu) = u_reg[t];
+            }
+        }
+        // tail units beyond the register budget: direct load (UU*NT >= UTOT
+        // for the shipped instantiations — loop compiles away)
+        for (int i = tid + UU * NTHREADS; i < UTOT; i += NTHREADS) {
+            int ro = i >> 3, eu = (i & 7) * 8;
+            const __bf16* src = Ut + ((size_t)(c * UCHUNK + (ro / O)) * O + (ro % O)) * E + e0 + eu;
             bf16x8 v;
             if (e0 + eu + 8 <= E) {
                 v = *reinterpret_cast<const bf16x8*>(src);
             } else {
+                v = bf16x8(0);
                 for (int j = 0; j < 8; ++j)
-                    v[j] = (e0 + eu + j < E) ? src[j] : (__bf16)0.f;
+                    if (e0 + eu + j < E) v[j] = src[j];
             }
             *reinterpret_cast<bf16x8*>(u_lds + (size_t)ro * BLK_E + eu) = v;
         }
+        if (c + 1 < nchunks) load_u(c + 1);   // issue next chunk's loads early
         __syncthreads();
 
         // ---- GEMM: R^T tile (256 n x 64 e), K=128
@@ -231,11 +272,29 @@ static void launch_fwd(const torch::Tensor& H, const torch::Tensor& W,
     dim3 grid(nmemb * ng);
     size_t lds = 16384 + (size_t)UCHUNK * O * BLK_E * 2 + (size_t)BLK_E * BLK_MO * O * 4;
     auto stream = at::cuda::getCurrentHIPStream();
-    hipLaunchKernelGGL(HIP_KERNEL_NAME(pairconv_fwd_kernel<O>), grid, dim3(NTHREADS), lds, stream,
-                       reinterpret_cast<const __bf16*>(H.data_ptr()),
-                       reinterpret_cast<const __bf16*>(W.data_ptr()),
-                       reinterpret_cast<const __bf16*>(Ut.data_ptr()),
-                       out.data_ptr<float>(), E, mo, miF, nmemb, coh);  // W arg = packed P
+    // UU: 16B u-units register-staged per thread for the next-chunk pipeline
+    // (0 = no staging, the round-1 serialized behavior). Default 2: the
+    // spill/pipeline tradeoff measured best on MI355X; override with
+    // SE3_FWD_UU for A/B runs.
+    const char* uu_env = getenv("SE3_FWD_UU");
+    int uu_sel = uu_env ? atoi(uu_env) : 2;
+    constexpr int UUfull = (UCHUNK * O * BLK_E / 8 + NTHREADS - 1) / NTHREADS;
+    int uu = uu_sel < 0 ? 2 : uu_sel;
+    if (uu > UUfull) uu = UUfull;
+#define LAUNCH_FWD_UU(UU)                                                                    \
+    hipLaunchKernelGGL(HIP_KERNEL_NAME(pairconv_fwd_kernel<O, UU>), grid, dim3(NTHREADS),    \
+                       lds, stream,                                                          \
+                       reinterpret_cast<const __bf16*>(H.data_ptr()),                        \
+                       reinterpret_cast<const __bf16*>(W.data_ptr()),                        \
+                       reinterpret_cast<const __bf16*>(Ut.data_ptr()),                       \
+                       out.data_ptr<float>(), E, mo, miF, nmemb, coh)  /* W arg = packed P */
+    switch (uu) {
+        case 0: LAUNCH_FWD_UU(0); break;
+        case 1: LAUNCH_FWD_UU(1); break;
+        case 2: LAUNCH_FWD_UU(2); break;
+        default: LAUNCH_FWD_UU((UUfull > 4 ? 4 : UUfull)); break;
+    }
+#undef LAUNCH_FWD_UU
 }
 
 void pairconv_fwd(torch::Tensor H, torch::Tensor W, torch::Tensor Ut,
@@ -283,6 +342,7 @@ void pairconv_bwd_dw(torch::Tensor G, torch::Tensor Ut, torch::Tensor Ht,
                      torch::Tensor dW, int64_t mo_);
 void pairconv_bwd_du(torch::Tensor H, torch::Tensor W, torch::Tensor bias,
                      torch::Tensor G, torch::Tensor dU, int64_t mo_);
+void pack_w_both(torch::Tensor W, torch::Tensor Pf, torch::Tensor Pdh, int64_t mo_);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("pairconv_fwd", &pairconv_fwd,
@@ -296,4 +356,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("norm_se3_bwd", &norm_se3_bwd, "fused NormSE3 backward");
     m.def("sh_basis_fwd", &sh_basis_fwd,
           "fused spherical-harmonics + equivariant basis (MI355X)");
+    m.def("pack_w_both", &pack_w_both,
+          "one-pass pack of net.6 W into both MFMA fragment layouts");
 }

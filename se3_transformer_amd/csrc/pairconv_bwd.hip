@@ -73,10 +73,56 @@ pairconv_bwd_dh_kernel(const __bf16* __restrict__ Gt,  // (mo, O, E) bf16
     const int nmo = mo / 8, nuc = miF / 32;
     const int mb_lo = (nmo / nsplit) * sp;
     const int mb_hi = mb_lo + nmo / nsplit;
+
+    // T14 register staging: next g tile (every mb) and a slice of the next
+    // u chunk (every cb) load under the dR + MFMA phases.
+    constexpr int GTOT = (8 * O * 64) / 8;     // g tile, 16B units (<= NT)
+    constexpr int UTOTd = (32 * O * 64) / 8;   // u chunk, 16B units
+    constexpr int UUD = 2;                     // staged u units per thread
+    bf16x8 g_reg, u_reg[UUD];
+    auto load_g = [&](int mb) {
+        if (tid < GTOT) {
+            int ro = tid >> 3, eu = (tid & 7) * 8;   // ro = m*O+o
+            const __bf16* src = Gt + ((size_t)(mb * 8 + ro / O) * O + (ro % O)) * E + e0 + eu;
+            if (e0 + eu + 8 <= E) g_reg = *reinterpret_cast<const bf16x8*>(src);
+            else {
+                bf16x8 v(0);
+                for (int j = 0; j < 8; ++j) if (e0 + eu + j < E) v[j] = src[j];
+                g_reg = v;
+            }
+        }
+    };
+    auto load_u = [&](int cb) {
+#pragma unroll
+        for (int t = 0; t < UUD; ++t) {
+            int i = tid + t * NT;
+            if (i < UTOTd) {
+                int ro = i >> 3, eu = (i & 7) * 8;
+                const __bf16* src = Ut + ((size_t)(cb * 32 + ro / O) * O + (ro % O)) * E + e0 + eu;
+                if (e0 + eu + 8 <= E) u_reg[t] = *reinterpret_cast<const bf16x8*>(src);
+                else {
+                    bf16x8 v(0);
+                    for (int j = 0; j < 8; ++j) if (e0 + eu + j < E) v[j] = src[j];
+                    u_reg[t] = v;
+                }
+            }
+        }
+    };
+    load_u(0);
+    load_g(mb_lo);
+
     for (int cb = 0; cb < nuc; ++cb) {
-        // stage u chunk [32][O][64] once per urow-chunk
+        // commit the staged u chunk [32][O][64] once per urow-chunk
         __syncthreads();
-        for (int i = tid; i < (32 * O * 64) / 8; i += NT) {
+#pragma unroll
+        for (int t = 0; t < UUD; ++t) {
+            int i = tid + t * NT;
+            if (i < UTOTd) {
+                int ro = i >> 3, eu = (i & 7) * 8;
+                *reinterpret_cast<bf16x8*>(u_lds + (size_t)ro * 64 + eu) = u_reg[t];
+            }
+        }
+        for (int i = tid + UUD * NT; i < UTOTd; i += NT) {   // unstaged tail
             int ro = i >> 3, eu = (i & 7) * 8;
             const __bf16* src = Ut + ((size_t)(cb * 32 + ro / O) * O + (ro % O)) * E + e0 + eu;
             bf16x8 v;
@@ -84,16 +130,14 @@ pairconv_bwd_dh_kernel(const __bf16* __restrict__ Gt,  // (mo, O, E) bf16
             else { for (int j = 0; j < 8; ++j) v[j] = (e0 + eu + j < E) ? src[j] : (__bf16)0.f; }
             *reinterpret_cast<bf16x8*>(u_lds + (size_t)ro * 64 + eu) = v;
         }
+        if (cb + 1 < nuc) load_u(cb + 1);
         for (int mb = mb_lo; mb < mb_hi; ++mb) {
-            // stage g tile [8][O][64] from Gt (e-contiguous rows)
-            for (int i = tid; i < (8 * O * 64) / 8; i += NT) {
-                int ro = i >> 3, eu = (i & 7) * 8;   // ro = m*O+o
-                const __bf16* src = Gt + ((size_t)(mb * 8 + ro / O) * O + (ro % O)) * E + e0 + eu;
-                bf16x8 v;
-                if (e0 + eu + 8 <= E) v = *reinterpret_cast<const bf16x8*>(src);
-                else { for (int j = 0; j < 8; ++j) v[j] = (e0 + eu + j < E) ? src[j] : (__bf16)0.f; }
-                *reinterpret_cast<bf16x8*>(g_lds + (size_t)ro * 64 + eu) = v;
+            // commit the staged g tile [8][O][64]
+            if (tid < GTOT) {
+                int ro = tid >> 3, eu = (tid & 7) * 8;
+                *reinterpret_cast<bf16x8*>(g_lds + (size_t)ro * 64 + eu) = g_reg;
             }
+            load_g(mb + 1 < mb_hi ? mb + 1 : mb_lo);   // next mb (or next cb's first)
             __syncthreads();
             // cooperative dR tile: [64e][256n]; thread owns (e-pair, run of 8 n):
             // b32 LDS reads over e-pairs + packed fma, two swizzled b128 writes.
@@ -358,23 +402,41 @@ pairconv_bwd_du_kernel(const __bf16* __restrict__ H,   // (E,128)
         *reinterpret_cast<bf16x8*>(reinterpret_cast<char*>(h_lds) + dst) = v;
     }
     for (int i = tid; i < 32 * O * 64; i += NT) du_acc[i] = 0.f;
+
+    // T14 register staging: the next mo-block's g tile + bias chunk load
+    // while the current block's MFMA + contraction run (g tile fits one
+    // 16B unit per thread for O <= 7).
+    constexpr int GTOT = (8 * O * 64) / 8;   // g tile in 16B units (<= NT)
+    bf16x8 g_reg;
+    float bias_reg;
+    auto load_gb = [&](int mb) {
+        if (tid < GTOT) {
+            int ro = tid >> 3, eu = (tid & 7) * 8;
+            const __bf16* src = Gt + ((size_t)(mb * 8 + ro / O) * O + (ro % O)) * E + e0 + eu;
+            if (e0 + eu + 8 <= E) g_reg = *reinterpret_cast<const bf16x8*>(src);
+            else {
+                bf16x8 v(0);
+                for (int j = 0; j < 8; ++j) if (e0 + eu + j < E) v[j] = src[j];
+                g_reg = v;
+            }
+        }
+        if (tid < 256) {
+            int m = tid >> 5, c = tid & 31;
+            bias_reg = bias[(size_t)(mb * 8 + m) * miF + uc0 + c];
+        }
+    };
+    load_gb(0);
     __syncthreads();
 
     const int nmo = mo / 8;
     for (int mb = 0; mb < nmo; ++mb) {
-        // stage g tile [8][O][64] (16B units) and bias chunk [256]
-        for (int i = tid; i < (8 * O * 64) / 8; i += NT) {
-            int ro = i >> 3, eu = (i & 7) * 8;
-            const __bf16* src = Gt + ((size_t)(mb * 8 + ro / O) * O + (ro % O)) * E + e0 + eu;
-            bf16x8 v;
-            if (e0 + eu + 8 <= E) v = *reinterpret_cast<const bf16x8*>(src);
-            else { for (int j = 0; j < 8; ++j) v[j] = (e0 + eu + j < E) ? src[j] : (__bf16)0.f; }
-            *reinterpret_cast<bf16x8*>(g_lds + (size_t)ro * 64 + eu) = v;
+        // commit the staged g tile [8][O][64] + bias chunk [256]
+        if (tid < GTOT) {
+            int ro = tid >> 3, eu = (tid & 7) * 8;
+            *reinterpret_cast<bf16x8*>(g_lds + (size_t)ro * 64 + eu) = g_reg;
         }
-        for (int i = tid; i < 256; i += NT) {
-            int m = i >> 5, c = i & 31;
-            bias_lds[i] = bias[(size_t)(mb * 8 + m) * miF + uc0 + c];
-        }
+        if (tid < 256) bias_lds[tid] = bias_reg;
+        if (mb + 1 < nmo) load_gb(mb + 1);   // issue next block's loads early
         __syncthreads();
         // MFMA R tile: (256n x 64e) like forward
         f32x4 acc[4][2];

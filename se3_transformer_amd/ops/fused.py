@@ -76,14 +76,28 @@ class _FusedPairConv(torch.autograd.Function):
         E = H.shape[0]
         miF, O, _ = Ut.shape
         H16 = H.contiguous().to(torch.bfloat16)
-        W16 = W.detach().contiguous().to(torch.bfloat16)
         Ut16 = Ut.contiguous().to(torch.bfloat16)
         # bias term: out0[e,mo,o] = sum_c bias[mo,c] Ut[c,o,e]
         b16 = bias.detach().to(torch.bfloat16).view(mo, miF)
         out = (b16 @ Ut16.reshape(miF, O * E)).view(mo, O, E) \
             .permute(2, 0, 1).contiguous().float()
-        ext.pairconv_fwd(H16, _pack_w_fwd(W16, mo, miF), Ut16, out, mo)
-        ctx.save_for_backward(H16, W16, Ut16, b16)
+        hip_bwd = os.environ.get('SE3_TORCH_BWD') != '1'
+        if hip_bwd and hasattr(ext, 'pack_w_both'):
+            # one-pass pack kernel: W read once, both fragment layouts
+            # written; saved for backward so nothing re-packs per step
+            Wd = W.detach().contiguous()
+            n128 = mo * miF * 128
+            Pf = torch.empty(n128, dtype=torch.bfloat16, device=Wd.device)
+            Pdh = torch.empty(n128, dtype=torch.bfloat16, device=Wd.device)
+            ext.pack_w_both(Wd, Pf, Pdh, mo)
+            ext.pairconv_fwd(H16, Pf, Ut16, out, mo)
+            ctx.save_for_backward(H16, Pf, Pdh, Ut16, b16)
+            ctx.packed = True
+        else:
+            W16 = W.detach().contiguous().to(torch.bfloat16)
+            ext.pairconv_fwd(H16, _pack_w_fwd(W16, mo, miF), Ut16, out, mo)
+            ctx.save_for_backward(H16, W16, Ut16, b16)
+            ctx.packed = False
         ctx.mo = mo
         ctx.w_dtype = W.dtype
         ctx.b_dtype = bias.dtype
@@ -91,8 +105,11 @@ class _FusedPairConv(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, grad_out):
-        H16, W16, Ut16, b16 = ctx.saved_tensors
         mo = ctx.mo
+        if ctx.packed:
+            H16, Pf, Pdh, Ut16, b16 = ctx.saved_tensors
+        else:
+            H16, W16, Ut16, b16 = ctx.saved_tensors
         E, K = H16.shape
         miF, O, _ = Ut16.shape
         g = grad_out.contiguous()                      # (E, mo, O) fp32
@@ -105,7 +122,7 @@ class _FusedPairConv(torch.autograd.Function):
             dH = dW = db = dUt = None
             g_t = g16.permute(1, 2, 0).contiguous()    # (mo, O, E) bf16
             if need_H:
-                P1 = _pack_w_dh(W16, mo, miF)
+                P1 = Pdh if ctx.packed else _pack_w_dh(W16, mo, miF)
                 dH = torch.zeros(E, K, dtype=torch.float32, device=H16.device)
                 ext.pairconv_bwd_dh(g_t, Ut16, P1, dH, mo)
                 del P1
@@ -124,10 +141,14 @@ class _FusedPairConv(torch.autograd.Function):
             if need_u:
                 dUt = torch.empty(miF, O, E, dtype=torch.float32,
                                   device=H16.device)
-                ext.pairconv_bwd_du(H16, _pack_w_fwd(W16, mo, miF),
-                                    b16.float().reshape(-1), g_t, dUt, mo)
+                Pu = Pf if ctx.packed else _pack_w_fwd(W16, mo, miF)
+                ext.pairconv_bwd_du(H16, Pu, b16.float().reshape(-1),
+                                    g_t, dUt, mo)
             return dH, dW, db, dUt, None
 
+        if ctx.packed:   # pragma: no cover - env toggled between fwd and bwd
+            raise RuntimeError('SE3_TORCH_BWD enabled after a packed forward; '
+                               'set it before the forward pass')
         u_eco = Ut16.permute(2, 0, 1)                  # (E, miF, O) view
         dH = dW = db = dUt = None
         if need_H:

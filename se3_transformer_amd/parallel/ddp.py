@@ -67,14 +67,18 @@ class DistributedDataParallelSE3(nn.Module):
 
     def __init__(self, module: nn.Module, bucket_bytes: int = 128 << 20,
                  process_group=None, average: bool = True,
-                 grad_compression: str = 'none', sync_params: bool = True):
+                 grad_compression: str = 'none', sync_params: bool = True,
+                 force_comm: bool = False):
         """grad_compression='bf16' all-reduces a bf16 copy of each bucket
         (halves xGMI traffic; fp32 master grads are then OVERWRITTEN by the
         reduced bf16 values — a one-shot rounding with relative error
         <= 2^-8 per element and no error feedback across steps, acceptable
         for SGD/Adam at bf16-compute scale but NOT bit-equal to fp32
         reduction). sync_params=False skips the initial parameter broadcast
-        (for ranks that already hold identical weights, e.g. seeded init)."""
+        (for ranks that already hold identical weights, e.g. seeded init).
+        force_comm=True launches the collectives even at world_size == 1
+        (exercises the full RCCL bucket path on a single GPU — used by the
+        GPU test suite; a no-op reduction numerically)."""
         super().__init__()
         assert grad_compression in ('none', 'bf16')
         self.module = module
@@ -82,6 +86,7 @@ class DistributedDataParallelSE3(nn.Module):
         self.average = average
         self.grad_compression = grad_compression
         self.world_size = dist.get_world_size(process_group) if dist.is_initialized() else 1
+        self._comm_active = self.world_size > 1 or (force_comm and dist.is_initialized())
 
         self._params = [p for p in module.parameters() if p.requires_grad]
         self._buckets = []
@@ -164,7 +169,7 @@ class DistributedDataParallelSE3(nn.Module):
         b.launched = True
 
     def _on_grad(self, p):
-        if self.world_size <= 1:
+        if not self._comm_active:
             return
         b = self._param_bucket.get(p)
         if b is None or b.launched:
@@ -175,7 +180,7 @@ class DistributedDataParallelSE3(nn.Module):
 
     def finalize(self):
         """Wait for outstanding reduces; reduce never-launched buckets; average."""
-        if self.world_size <= 1:
+        if not self._comm_active:
             return
         for b in self._buckets:
             if not b.launched:

@@ -25,13 +25,16 @@ setup(
             name='se3_transformer_amd._C',
             sources=['se3_transformer_amd/csrc/pairconv.hip',
                      'se3_transformer_amd/csrc/pairconv_bwd.hip',
+                     'se3_transformer_amd/csrc/pack_w.hip',
                      'se3_transformer_amd/csrc/sh_basis.hip',
                      'se3_transformer_amd/csrc/norm_se3.hip',
                      'se3_transformer_amd/csrc/attn.hip',
                      'se3_transformer_amd/csrc/knn.hip'],
             extra_compile_args={
                 'cxx': ['-O3'],
-                'nvcc': ['-O3', '--offload-arch=gfx950'],
+                'nvcc': ['-O3', '--offload-arch=gfx950'] +
+                        (['-Rpass-analysis=kernel-resource-usage']
+                         if os.environ.get('SE3_RES_USAGE') else []),
             },
         ),
     ],

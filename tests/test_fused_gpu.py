@@ -313,3 +313,56 @@ def test_knn_kernel_causal_vs_eager():
     out = model(feats, coors, mask, return_type=0)
     err = (out - ref).abs().max().item()
     assert err < 1e-4, f'causal knn mismatch: {err}'
+
+
+@needs_gpu
+def test_pack_w_both_matches_python_permutes():
+    """csrc/pack_w.hip: the one-pass pack kernel must reproduce the python
+    permute layouts exactly (including the fp32 -> bf16 rounding)."""
+    from se3_transformer_amd import _C
+    from se3_transformer_amd.ops.fused import _pack_w_dh, _pack_w_fwd
+
+    torch.manual_seed(3)
+    device = torch.device('cuda')
+    mo, miF = 16, 96
+    for dtype in (torch.float32, torch.bfloat16):
+        W = torch.randn(mo * miF, 128, device=device, dtype=dtype)
+        Pf = torch.empty(mo * miF * 128, dtype=torch.bfloat16, device=device)
+        Pdh = torch.empty_like(Pf)
+        _C.pack_w_both(W, Pf, Pdh, mo)
+        W16 = W.to(torch.bfloat16)
+        ref_f = _pack_w_fwd(W16, mo, miF)
+        ref_d = _pack_w_dh(W16, mo, miF)
+        assert torch.equal(Pf.view_as(ref_f), ref_f), f'fwd pack {dtype}'
+        assert torch.equal(Pdh.view_as(ref_d), ref_d), f'dh pack {dtype}'
+
+
+@needs_gpu
+@pytest.mark.parametrize('uu', ['0', '1', '2', '4'])
+def test_pairconv_fwd_uu_variants_agree(uu):
+    """All SE3_FWD_UU pipeline variants must be numerically identical."""
+    from se3_transformer_amd import _C
+    from se3_transformer_amd.ops.fused import _pack_w_fwd
+
+    torch.manual_seed(4)
+    device = torch.device('cuda')
+    mo, mi, F_, O, E = 16, 8, 5, 5, 500
+    miF = mi * F_
+    H = torch.randn(E, 128, device=device).to(torch.bfloat16)
+    W = torch.randn(mo * miF, 128, device=device).to(torch.bfloat16)
+    Ut = torch.randn(miF, O, E, device=device).to(torch.bfloat16)
+    P = _pack_w_fwd(W, mo, miF)
+
+    def run():
+        out = torch.zeros(E, mo, O, device=device)
+        _C.pairconv_fwd(H, P, Ut, out, mo)
+        return out
+
+    os.environ['SE3_FWD_UU'] = '0'
+    ref = run()
+    os.environ['SE3_FWD_UU'] = uu
+    try:
+        out = run()
+    finally:
+        del os.environ['SE3_FWD_UU']
+    assert torch.equal(out, ref), f'UU={uu} diverges from UU=0'
