@@ -272,12 +272,14 @@ static void launch_fwd(const torch::Tensor& H, const torch::Tensor& W,
     dim3 grid(nmemb * ng);
     size_t lds = 16384 + (size_t)UCHUNK * O * BLK_E * 2 + (size_t)BLK_E * BLK_MO * O * 4;
     auto stream = at::cuda::getCurrentHIPStream();
-    // UU: 16B u-units register-staged per thread for the next-chunk pipeline
-    // (0 = no staging, the round-1 serialized behavior). Default 2: the
-    // spill/pipeline tradeoff measured best on MI355X; override with
-    // SE3_FWD_UU for A/B runs.
+    // UU: 16B u-units register-staged per thread for the next-chunk pipeline.
+    // MEASURED NEGATIVE on MI355X for the hot O=7 shape (UU=0: 239.6 TF/s,
+    // UU=1: 217, UU=2: 221, UU=4: 153 — the VGPR spill the staging forces
+    // under the 2-blocks/CU cap costs more than the hidden u latency, and
+    // the co-resident block already covers most of the stall). Default 0;
+    // SE3_FWD_UU kept for A/B runs.
     const char* uu_env = getenv("SE3_FWD_UU");
-    int uu_sel = uu_env ? atoi(uu_env) : 2;
+    int uu_sel = uu_env ? atoi(uu_env) : 0;
     constexpr int UUfull = (UCHUNK * O * BLK_E / 8 + NTHREADS - 1) / NTHREADS;
     int uu = uu_sel < 0 ? 2 : uu_sel;
     if (uu > UUfull) uu = UUfull;

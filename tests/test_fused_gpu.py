@@ -346,8 +346,8 @@ def test_pairconv_fwd_uu_variants_agree(uu):
 
     torch.manual_seed(4)
     device = torch.device('cuda')
-    mo, mi, F_, O, E = 16, 8, 5, 5, 500
-    miF = mi * F_
+    mo, mi, F_, O, E = 16, 32, 5, 5, 500
+    miF = mi * F_   # 160: must be a multiple of 32 (kernel contract)
     H = torch.randn(E, 128, device=device).to(torch.bfloat16)
     W = torch.randn(mo * miF, 128, device=device).to(torch.bfloat16)
     Ut = torch.randn(miF, O, E, device=device).to(torch.bfloat16)

@@ -96,16 +96,16 @@ def test_ddp_bucket_path_over_rccl(nccl_pg, compression):
     # with >1 bucket, at least one all-reduce must have been launched by the
     # grad hooks before finalize (the overlap-with-backward mechanism)
     assert len(ddp._buckets) > 1 and launched_during_backward >= 1
-    tol = 0.0 if compression == 'none' else 1e-2
+    # gather-backward uses atomicAdd on GPU (accumulation-order
+    # nondeterminism), so even the uncompressed path is compared with a
+    # tight tolerance rather than bit-equality
+    tol = 1e-5 if compression == 'none' else 1e-2
     for n, p in model.named_parameters():
         if n not in ref:
             continue
-        if tol == 0.0:
-            assert torch.equal(p.grad, ref[n]), f'grad mismatch {n}'
-        else:
-            denom = ref[n].abs().max().clamp(min=1e-6)
-            assert ((p.grad - ref[n]).abs().max() / denom) < tol, \
-                f'bf16-compressed grad off {n}'
+        denom = ref[n].abs().max().clamp(min=1e-6)
+        err = ((p.grad - ref[n]).abs().max() / denom).item()
+        assert err < tol, f'grad mismatch {n}: rel {err}'
 
 
 @needs_gpu
