@@ -68,12 +68,15 @@ def main():
               f'parity fwd={okf} dh={okd}')
 
     if os.environ.get('SE3_SWEEP_UU'):
-        for uu in ('0', '1', '2', '4'):
+        for uu, mb2 in (('0', '1'), ('1', '1'), ('2', '1'),
+                        ('0', '2'), ('1', '2'), ('2', '2')):
             os.environ['SE3_FWD_UU'] = uu
+            os.environ['SE3_FWD_MB2'] = mb2
             ms = timeit(lambda: _C.pairconv_fwd(H, P, Ut, out, mo))
-            print(f'fwd UU={uu} ({di},{do}): {ms:8.3f} ms  '
+            print(f'fwd UU={uu} MB2={mb2} ({di},{do}): {ms:8.3f} ms  '
                   f'{gemm_fl/ms/1e9:7.1f} TF/s (gemm)')
         del os.environ['SE3_FWD_UU']
+        del os.environ['SE3_FWD_MB2']
     if args.only:
         fn = {'fwd': lambda: _C.pairconv_fwd(H, P, Ut, out, mo),
               'dh': lambda: _C.pairconv_bwd_dh(gt, Ut, _pack_w_dh(W, mo, miF),

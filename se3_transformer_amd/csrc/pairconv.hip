@@ -51,7 +51,12 @@ __device__ __forceinline__ float bf16_to_f32(unsigned short u) {
     return v.f;
 }
 
-template <int O, int UU>   // UU = 16B u-units register-staged per thread
+// MB2 = mo-blocks (of 8) processed per workgroup against ONE staged u chunk.
+// MB2=2 halves the u-chunk HBM traffic (u is re-read per mo-block otherwise)
+// while keeping 2-blocks/CU residency: the two sub-blocks run serially, so
+// the MFMA accumulator registers are reused, only the LDS `part` doubles
+// (O=7: 16K h + 28K u + 28K part = 72 KiB <= 80 KiB per block).
+template <int O, int UU, int MB2>   // UU = 16B u-units register-staged per thread
 __global__ void __launch_bounds__(NTHREADS, 4)   // cap VGPR<=128: 2 blocks/CU
 pairconv_fwd_kernel(const __bf16* __restrict__ H,
                     const __bf16* __restrict__ P,   // packed W: [mo/8][miF/32][wm4][mf4][kit4][lane64][8]
@@ -62,7 +67,7 @@ pairconv_fwd_kernel(const __bf16* __restrict__ H,
     // carve: H tile | u chunk | partial accumulator
     __bf16* h_lds = reinterpret_cast<__bf16*>(smem);                       // [64][128] swizzled, 16 KiB
     __bf16* u_lds = reinterpret_cast<__bf16*>(smem + 16384);               // [32][O][64]
-    float* part = reinterpret_cast<float*>(smem + 16384 + UCHUNK * O * BLK_E * 2); // [64][8][O]
+    float* part = reinterpret_cast<float*>(smem + 16384 + UCHUNK * O * BLK_E * 2); // [64][8*MB2][O]
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -84,7 +89,7 @@ pairconv_fwd_kernel(const __bf16* __restrict__ H,
         mb = blockIdx.x / nmemb;
     }
     const int e0 = eb * BLK_E;
-    const int mo0 = mb * BLK_MO;
+    const int mo0 = mb * BLK_MO * MB2;
 
     // ---- stage H tile (64 x 128 bf16), XOR-swizzled 16B slots within each row
     {
@@ -101,7 +106,7 @@ pairconv_fwd_kernel(const __bf16* __restrict__ H,
             *reinterpret_cast<bf16x8*>(reinterpret_cast<char*>(h_lds) + dst) = v;
         }
         // zero partial accumulator
-        for (int i = tid; i < BLK_E * BLK_MO * O; i += NTHREADS) part[i] = 0.f;
+        for (int i = tid; i < BLK_E * BLK_MO * MB2 * O; i += NTHREADS) part[i] = 0.f;
     }
 
     constexpr int UTOT = (UCHUNK * O * BLK_E) / 8;   // u chunk in 16B units
@@ -159,6 +164,13 @@ pairconv_fwd_kernel(const __bf16* __restrict__ H,
         if (c + 1 < nchunks) load_u(c + 1);   // issue next chunk's loads early
         __syncthreads();
 
+        // ---- per mo-sub-block: GEMM + epilogue against the SAME u chunk.
+        // Unrolled, but a sched_barrier between the sub-blocks keeps the
+        // second sub-block's W-fragment loads from being hoisted into the
+        // first (which would double the live A-operand registers and spill).
+#pragma unroll
+        for (int sub = 0; sub < MB2; ++sub) {
+        if (MB2 > 1 && sub > 0) __builtin_amdgcn_sched_barrier(0);
         // ---- GEMM: R^T tile (256 n x 64 e), K=128
         f32x4 acc[4][2];
 #pragma unroll
@@ -166,7 +178,7 @@ pairconv_fwd_kernel(const __bf16* __restrict__ H,
 #pragma unroll
             for (int ef = 0; ef < 2; ++ef) acc[mf][ef] = f32x4(0.f);
 
-        const __bf16* pbase = P + ((((size_t)mb * (miF / 32) + c) * 4 + wm) * 4) * 4 * 64 * 8
+        const __bf16* pbase = P + ((((size_t)(mb * MB2 + sub) * (miF / 32) + c) * 4 + wm) * 4) * 4 * 64 * 8
                               + (size_t)lane * 8;
 #pragma unroll
         for (int kit = 0; kit < 4; ++kit) {
@@ -238,26 +250,27 @@ pairconv_fwd_kernel(const __bf16* __restrict__ H,
                 const int e = we * 32 + ef * 16 + l15;
 #pragma unroll
                 for (int mi_ = 0; mi_ < 2; ++mi_) {
-                    const int moi = wm * 2 + mi_;
+                    const int moi = sub * BLK_MO + wm * 2 + mi_;
 #pragma unroll
                     for (int o = 0; o < O; ++o) {
-                        float* p = part + ((size_t)e * BLK_MO + moi) * O + o;
+                        float* p = part + ((size_t)e * (BLK_MO * MB2) + moi) * O + o;
                         *p += s[ef][mi_][o];
                     }
                 }
             }
         }
+        }   // sub
         __syncthreads();
     }
 
     // ---- write out: out[e0+e][mo0+moi][o] += partial
-    for (int i = tid; i < BLK_E * BLK_MO * O; i += NTHREADS) {
+    for (int i = tid; i < BLK_E * BLK_MO * MB2 * O; i += NTHREADS) {
         int o = i % O;
-        int moi = (i / O) % BLK_MO;
-        int e = i / (O * BLK_MO);
+        int moi = (i / O) % (BLK_MO * MB2);
+        int e = i / (O * BLK_MO * MB2);
         if (e0 + e < E) {
             float* p = out + ((size_t)(e0 + e) * mo + mo0 + moi) * O + o;
-            *p += part[((size_t)e * BLK_MO + moi) * O + o];
+            *p += part[((size_t)e * (BLK_MO * MB2) + moi) * O + o];
         }
     }
 }
@@ -267,10 +280,6 @@ static void launch_fwd(const torch::Tensor& H, const torch::Tensor& W,
                        const torch::Tensor& Ut, torch::Tensor& out,
                        int E, int mo, int miF) {
     int nmemb = (E + BLK_E - 1) / BLK_E;
-    int ng = mo / BLK_MO;
-    int coh = (ng % 8 == 0) ? 1 : 0;
-    dim3 grid(nmemb * ng);
-    size_t lds = 16384 + (size_t)UCHUNK * O * BLK_E * 2 + (size_t)BLK_E * BLK_MO * O * 4;
     auto stream = at::cuda::getCurrentHIPStream();
     // UU: 16B u-units register-staged per thread for the next-chunk pipeline.
     // MEASURED NEGATIVE on MI355X for the hot O=7 shape (UU=0: 239.6 TF/s,
@@ -281,22 +290,42 @@ static void launch_fwd(const torch::Tensor& H, const torch::Tensor& W,
     const char* uu_env = getenv("SE3_FWD_UU");
     int uu_sel = uu_env ? atoi(uu_env) : 0;
     constexpr int UUfull = (UCHUNK * O * BLK_E / 8 + NTHREADS - 1) / NTHREADS;
-    int uu = uu_sel < 0 ? 2 : uu_sel;
+    int uu = uu_sel < 0 ? 0 : uu_sel;
+    if (uu > 2) uu = 2;
     if (uu > UUfull) uu = UUfull;
-#define LAUNCH_FWD_UU(UU)                                                                    \
-    hipLaunchKernelGGL(HIP_KERNEL_NAME(pairconv_fwd_kernel<O, UU>), grid, dim3(NTHREADS),    \
-                       lds, stream,                                                          \
+    // MB2: mo-blocks per workgroup sharing one staged u chunk (halves u
+    // traffic). Default 2 where mo allows; SE3_FWD_MB2=1 forces the old
+    // one-block grid for A/B runs.
+    const char* mb_env = getenv("SE3_FWD_MB2");
+    int mb2 = (mb_env ? atoi(mb_env) : 2);
+    if (mb2 != 1 && mo % (BLK_MO * 2) != 0) mb2 = 1;
+    if (mb2 != 1) mb2 = 2;
+    int ng = mo / (BLK_MO * mb2);
+    int coh = (ng % 8 == 0) ? 1 : 0;
+    dim3 grid(nmemb * ng);
+    size_t lds = 16384 + (size_t)UCHUNK * O * BLK_E * 2
+                 + (size_t)BLK_E * BLK_MO * mb2 * O * 4;
+#define LAUNCH_FWD(UU, MB2)                                                                  \
+    hipLaunchKernelGGL(HIP_KERNEL_NAME(pairconv_fwd_kernel<O, UU, MB2>), grid,               \
+                       dim3(NTHREADS), lds, stream,                                          \
                        reinterpret_cast<const __bf16*>(H.data_ptr()),                        \
                        reinterpret_cast<const __bf16*>(W.data_ptr()),                        \
                        reinterpret_cast<const __bf16*>(Ut.data_ptr()),                       \
                        out.data_ptr<float>(), E, mo, miF, nmemb, coh)  /* W arg = packed P */
-    switch (uu) {
-        case 0: LAUNCH_FWD_UU(0); break;
-        case 1: LAUNCH_FWD_UU(1); break;
-        case 2: LAUNCH_FWD_UU(2); break;
-        default: LAUNCH_FWD_UU((UUfull > 4 ? 4 : UUfull)); break;
+    if (mb2 == 2) {
+        switch (uu) {
+            case 0: LAUNCH_FWD(0, 2); break;
+            case 1: LAUNCH_FWD(1, 2); break;
+            default: LAUNCH_FWD(2, 2); break;
+        }
+    } else {
+        switch (uu) {
+            case 0: LAUNCH_FWD(0, 1); break;
+            case 1: LAUNCH_FWD(1, 1); break;
+            default: LAUNCH_FWD(2, 1); break;
+        }
     }
-#undef LAUNCH_FWD_UU
+#undef LAUNCH_FWD
 }
 
 void pairconv_fwd(torch::Tensor H, torch::Tensor W, torch::Tensor Ut,

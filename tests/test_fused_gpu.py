@@ -338,9 +338,10 @@ def test_pack_w_both_matches_python_permutes():
 
 
 @needs_gpu
-@pytest.mark.parametrize('uu', ['0', '1', '2', '4'])
-def test_pairconv_fwd_uu_variants_agree(uu):
-    """All SE3_FWD_UU pipeline variants must be numerically identical."""
+@pytest.mark.parametrize('uu,mb2', [('0', '1'), ('1', '1'), ('2', '1'),
+                                    ('0', '2'), ('2', '2')])
+def test_pairconv_fwd_uu_variants_agree(uu, mb2):
+    """All SE3_FWD_UU / SE3_FWD_MB2 variants must be numerically identical."""
     from se3_transformer_amd import _C
     from se3_transformer_amd.ops.fused import _pack_w_fwd
 
@@ -359,10 +360,13 @@ def test_pairconv_fwd_uu_variants_agree(uu):
         return out
 
     os.environ['SE3_FWD_UU'] = '0'
+    os.environ['SE3_FWD_MB2'] = '1'
     ref = run()
     os.environ['SE3_FWD_UU'] = uu
+    os.environ['SE3_FWD_MB2'] = mb2
     try:
         out = run()
     finally:
         del os.environ['SE3_FWD_UU']
-    assert torch.equal(out, ref), f'UU={uu} diverges from UU=0'
+        del os.environ['SE3_FWD_MB2']
+    assert torch.equal(out, ref), f'UU={uu} MB2={mb2} diverges'
