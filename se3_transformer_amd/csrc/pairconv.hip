@@ -374,6 +374,16 @@ void pairconv_bwd_dw(torch::Tensor G, torch::Tensor Ut, torch::Tensor Ht,
 void pairconv_bwd_du(torch::Tensor H, torch::Tensor W, torch::Tensor bias,
                      torch::Tensor G, torch::Tensor dU, int64_t mo_);
 void pack_w_both(torch::Tensor W, torch::Tensor Pf, torch::Tensor Pdh, int64_t mo_);
+void radial_trunk_fwd(torch::Tensor X, torch::Tensor W0, torch::Tensor p0,
+                      torch::Tensor W3, torch::Tensor p3,
+                      torch::Tensor H, torch::Tensor yh0, torch::Tensor yh3,
+                      torch::Tensor rs01, double eps);
+void radial_trunk_bwd(torch::Tensor dH, torch::Tensor X, torch::Tensor W0,
+                      torch::Tensor p0, torch::Tensor W3, torch::Tensor W3t,
+                      torch::Tensor p3, torch::Tensor yh0, torch::Tensor yh3,
+                      torch::Tensor rs01, torch::Tensor dW0, torch::Tensor dp0,
+                      torch::Tensor dW3, torch::Tensor dp3, torch::Tensor dX,
+                      double eps);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("pairconv_fwd", &pairconv_fwd,
@@ -389,4 +399,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "fused spherical-harmonics + equivariant basis (MI355X)");
     m.def("pack_w_both", &pack_w_both,
           "one-pass pack of net.6 W into both MFMA fragment layouts");
+    m.def("radial_trunk_fwd", &radial_trunk_fwd,
+          "fused radial trunk (Linear-LN-GELU x2) forward");
+    m.def("radial_trunk_bwd", &radial_trunk_bwd,
+          "fused radial trunk backward");
 }

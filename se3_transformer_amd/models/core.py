@@ -155,7 +155,23 @@ class RadialFunc(nn.Module):
 
     def hidden(self, x):
         """The 128-dim trunk activations (everything but the final Linear) —
-        used by the fused HIP path, which folds net.6 into the conv kernel."""
+        used by the fused HIP path, which folds net.6 into the conv kernel.
+
+        On CUDA(ROCm) the whole trunk (net.0 Linear + net.1 LN + GELU +
+        net.3 Linear + net.4 LN + GELU, reference :287-295) runs as ONE
+        HIP kernel each way (csrc/radial.hip); eager fallback otherwise."""
+        from ..ops import fused as _fused
+        in_dim = x.shape[-1]
+        if (x.is_cuda and os.environ.get('SE3_EAGER_RADIAL') != '1'
+                and _fused.radial_trunk_ok(in_dim, self.mid_dim)):
+            n0, ln1 = self.net[0], self.net[1]
+            n3, ln4 = self.net[3], self.net[4]
+            lead = x.shape[:-1]
+            h = _fused.radial_trunk(x.reshape(-1, in_dim),
+                                    n0.weight, n0.bias, ln1.weight, ln1.bias,
+                                    n3.weight, n3.bias, ln4.weight, ln4.bias,
+                                    ln1.eps)
+            return h.view(*lead, self.mid_dim)
         h = x
         for layer in self.net[:-1]:
             h = layer(h)

@@ -284,3 +284,64 @@ class _AttnFn(torch.autograd.Function):
 
 def fused_attention(q, k, v, mask_u8, n, heads, scale, kv_one=False):
     return _AttnFn.apply(q, k, v, mask_u8, n, heads, scale, kv_one)
+
+
+RADIAL_TRUNK_DIMS = (1, 2, 3, 9, 17)   # instantiated in csrc/radial.hip
+
+
+def radial_trunk_ok(in_dim: int, mid_dim: int) -> bool:
+    ext = _load_ext()
+    return (mid_dim == 128 and in_dim in RADIAL_TRUNK_DIMS
+            and ext is not None and hasattr(ext, 'radial_trunk_fwd'))
+
+
+class _RadialTrunkFn(torch.autograd.Function):
+    """Fused RadialFunc trunk (csrc/radial.hip): Linear(D->128) + LN + GELU
+    + Linear(128->128) + LN + GELU in one kernel each way. Mirrors the
+    autocast-bf16 eager numerics (GEMMs bf16/fp32-acc, LN+GELU fp32)."""
+
+    @staticmethod
+    def forward(ctx, x, w0, b0, g0, be0, w3, b3, g3, be3, eps):
+        ext = _load_ext()
+        E, D = x.shape
+        x16 = x.contiguous().to(torch.bfloat16)
+        w0_16 = w0.detach().contiguous().to(torch.bfloat16)
+        w3_16 = w3.detach().contiguous().to(torch.bfloat16)
+        p0 = torch.cat([b0.detach().float(), g0.detach().float(),
+                        be0.detach().float()]).contiguous()
+        p3 = torch.cat([b3.detach().float(), g3.detach().float(),
+                        be3.detach().float()]).contiguous()
+        H = torch.empty(E, 128, dtype=torch.bfloat16, device=x.device)
+        yh0 = torch.empty_like(H)
+        yh3 = torch.empty_like(H)
+        rs = torch.empty(2, E, dtype=torch.float32, device=x.device)
+        ext.radial_trunk_fwd(x16, w0_16, p0, w3_16, p3, H, yh0, yh3, rs, eps)
+        ctx.save_for_backward(x16, w0_16, p0, w3_16, p3, yh0, yh3, rs)
+        ctx.eps = eps
+        ctx.x_dtype = x.dtype
+        return H
+
+    @staticmethod
+    def backward(ctx, dH):
+        ext = _load_ext()
+        x16, w0_16, p0, w3_16, p3, yh0, yh3, rs = ctx.saved_tensors
+        E, D = x16.shape
+        dev = x16.device
+        dH16 = dH.contiguous().to(torch.bfloat16)
+        dW0 = torch.zeros(128, D, dtype=torch.float32, device=dev)
+        dp0 = torch.zeros(3 * 128, dtype=torch.float32, device=dev)
+        dW3 = torch.zeros(128, 128, dtype=torch.float32, device=dev)
+        dp3 = torch.zeros(3 * 128, dtype=torch.float32, device=dev)
+        need_x = ctx.needs_input_grad[0]
+        dX = (torch.zeros(E, D, dtype=torch.float32, device=dev)
+              if need_x else torch.empty(0, device=dev))
+        w3t = w3_16.t().contiguous()
+        ext.radial_trunk_bwd(dH16, x16, w0_16, p0, w3_16, w3t, p3,
+                             yh0, yh3, rs, dW0, dp0, dW3, dp3, dX, ctx.eps)
+        return ((dX.to(ctx.x_dtype) if need_x else None),
+                dW0, dp0[:128], dp0[128:256], dp0[256:],
+                dW3, dp3[:128], dp3[128:256], dp3[256:], None)
+
+
+def radial_trunk(x, w0, b0, g0, be0, w3, b3, g3, be3, eps=1e-5):
+    return _RadialTrunkFn.apply(x, w0, b0, g0, be0, w3, b3, g3, be3, eps)
