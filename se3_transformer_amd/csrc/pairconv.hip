@@ -64,10 +64,14 @@ pairconv_fwd_kernel(const __bf16* __restrict__ H,
                     float* __restrict__ out,
                     int E, int mo, int miF, int nmemb, int coh) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
-    // carve: H tile | u chunk | partial accumulator
+    // carve: H tile | u chunk | partial accumulator. The u chunk is stored
+    // [urow][e][o] with o PADDED to 8 so the epilogue reads one 16B vector
+    // per (urow, e) instead of O scalar ds_read_u16s — the epilogue VALU/LDS
+    // issue rate was co-limiting with the MFMA pipe (measured ~240 TF/s vs
+    // the ~380 TF/s W-stream roofline).
     __bf16* h_lds = reinterpret_cast<__bf16*>(smem);                       // [64][128] swizzled, 16 KiB
-    __bf16* u_lds = reinterpret_cast<__bf16*>(smem + 16384);               // [32][O][64]
-    float* part = reinterpret_cast<float*>(smem + 16384 + UCHUNK * O * BLK_E * 2); // [64][8*MB2][O]
+    __bf16* u_lds = reinterpret_cast<__bf16*>(smem + 16384);               // [32][64][8] 32 KiB
+    float* part = reinterpret_cast<float*>(smem + 16384 + UCHUNK * BLK_E * 8 * 2); // [64][8*MB2][O]
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -105,8 +109,11 @@ pairconv_fwd_kernel(const __bf16* __restrict__ H,
             }
             *reinterpret_cast<bf16x8*>(reinterpret_cast<char*>(h_lds) + dst) = v;
         }
-        // zero partial accumulator
+        // zero partial accumulator + the u pad lanes (o in [O, 8) are never
+        // overwritten by staging, so one upfront clear keeps them zero)
         for (int i = tid; i < BLK_E * BLK_MO * MB2 * O; i += NTHREADS) part[i] = 0.f;
+        for (int i = tid; i < UCHUNK * BLK_E * 8 / 8; i += NTHREADS)
+            *reinterpret_cast<bf16x8*>(u_lds + (size_t)i * 8) = bf16x8(0);
     }
 
     constexpr int UTOT = (UCHUNK * O * BLK_E) / 8;   // u chunk in 16B units
@@ -137,13 +144,16 @@ pairconv_fwd_kernel(const __bf16* __restrict__ H,
 
     const int nchunks = miF / UCHUNK;
     for (int c = 0; c < nchunks; ++c) {
-        // ---- commit the staged u chunk: u_lds[urow][o][e]
+        // ---- commit the staged u chunk, scattered to [urow][e][o(pad 8)]
 #pragma unroll
         for (int t = 0; t < UU; ++t) {
             int i = tid + t * NTHREADS;
             if (i < UTOT) {
                 int ro = i >> 3, eu = (i & 7) * 8;
-                *reinterpret_cast<bf16x8*>(u_lds + (size_t)ro * BLK_E + eu) = u_reg[t];
+                const int ur = ro / O, o = ro % O;
+#pragma unroll
+                for (int j = 0; j < 8; ++j)
+                    u_lds[((size_t)ur * BLK_E + eu + j) * 8 + o] = u_reg[t][j];
             }
         }
         // tail units beyond the register budget: direct load (UU*NT >= UTOT
@@ -159,7 +169,10 @@ pairconv_fwd_kernel(const __bf16* __restrict__ H,
                 for (int j = 0; j < 8; ++j)
                     if (e0 + eu + j < E) v[j] = src[j];
             }
-            *reinterpret_cast<bf16x8*>(u_lds + (size_t)ro * BLK_E + eu) = v;
+            const int ur = ro / O, o = ro % O;
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+                u_lds[((size_t)ur * BLK_E + eu + j) * 8 + o] = v[j];
         }
         if (c + 1 < nchunks) load_u(c + 1);   // issue next chunk's loads early
         __syncthreads();
@@ -201,14 +214,17 @@ pairconv_fwd_kernel(const __bf16* __restrict__ H,
                     acc[mf][ef] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[mf], b[ef], acc[mf][ef], 0, 0, 0);
         }
 
-        // ---- epilogue: contract acc against u_lds into s[ef][moi][o]
-        float s[2][2][O];
+        // ---- epilogue: contract acc against u_lds into s[ef][moi][o-pairs].
+        // One ds_read_b128 per (row, e) covers all 8 padded o; the o-pair
+        // float2 accumulation maps onto v_pk_fma_f32.
+        typedef __attribute__((ext_vector_type(2))) float f32x2e;
+        f32x2e s[2][2][4];
 #pragma unroll
         for (int ef = 0; ef < 2; ++ef)
 #pragma unroll
             for (int mi_ = 0; mi_ < 2; ++mi_)
 #pragma unroll
-                for (int o = 0; o < O; ++o) s[ef][mi_][o] = 0.f;
+                for (int p_ = 0; p_ < 4; ++p_) s[ef][mi_][p_] = f32x2e{0.f, 0.f};
 
 #pragma unroll
         for (int mf = 0; mf < 2; ++mf) {   // mf and mf+2 share urow (rows r, r+32)
@@ -221,12 +237,13 @@ pairconv_fwd_kernel(const __bf16* __restrict__ H,
                     const int e = we * 32 + ef * 16 + l15;
                     const float rv0 = acc[mf][ef][reg];
                     const float rv1 = acc[mf + 2][ef][reg];
+                    bf16x8 uv8 = *reinterpret_cast<const bf16x8*>(
+                        u_lds + ((size_t)urow * BLK_E + e) * 8);
 #pragma unroll
-                    for (int o = 0; o < O; ++o) {
-                        float uv = bf16_to_f32(
-                            reinterpret_cast<const unsigned short*>(u_lds)[(urow * O + o) * BLK_E + e]);
-                        s[ef][0][o] = fmaf(rv0, uv, s[ef][0][o]);
-                        s[ef][1][o] = fmaf(rv1, uv, s[ef][1][o]);
+                    for (int p_ = 0; p_ < 4; ++p_) {
+                        f32x2e u2 = {(float)uv8[2 * p_], (float)uv8[2 * p_ + 1]};
+                        s[ef][0][p_] += rv0 * u2;
+                        s[ef][1][p_] += rv1 * u2;
                     }
                 }
             }
@@ -238,11 +255,15 @@ pairconv_fwd_kernel(const __bf16* __restrict__ H,
 #pragma unroll
             for (int mi_ = 0; mi_ < 2; ++mi_)
 #pragma unroll
-                for (int o = 0; o < O; ++o) {
-                    float v = s[ef][mi_][o];
-                    v += __shfl_xor(v, 16);
-                    v += __shfl_xor(v, 32);
-                    s[ef][mi_][o] = v;
+                for (int p_ = 0; p_ < 4; ++p_) {
+#pragma unroll
+                    for (int c2 = 0; c2 < 2; ++c2) {
+                        if (2 * p_ + c2 >= O) break;
+                        float v = s[ef][mi_][p_][c2];
+                        v += __shfl_xor(v, 16);
+                        v += __shfl_xor(v, 32);
+                        s[ef][mi_][p_][c2] = v;
+                    }
                 }
         if (l4 == 0) {
 #pragma unroll
@@ -254,7 +275,7 @@ pairconv_fwd_kernel(const __bf16* __restrict__ H,
 #pragma unroll
                     for (int o = 0; o < O; ++o) {
                         float* p = part + ((size_t)e * (BLK_MO * MB2) + moi) * O + o;
-                        *p += s[ef][mi_][o];
+                        *p += s[ef][mi_][o >> 1][o & 1];
                     }
                 }
             }
@@ -303,7 +324,7 @@ static void launch_fwd(const torch::Tensor& H, const torch::Tensor& W,
     int ng = mo / (BLK_MO * mb2);
     int coh = (ng % 8 == 0) ? 1 : 0;
     dim3 grid(nmemb * ng);
-    size_t lds = 16384 + (size_t)UCHUNK * O * BLK_E * 2
+    size_t lds = 16384 + (size_t)UCHUNK * BLK_E * 8 * 2
                  + (size_t)BLK_E * BLK_MO * mb2 * O * 4;
 #define LAUNCH_FWD(UU, MB2)                                                                  \
     hipLaunchKernelGGL(HIP_KERNEL_NAME(pairconv_fwd_kernel<O, UU, MB2>), grid,               \
@@ -361,9 +382,17 @@ void sh_basis_fwd(torch::Tensor rel, torch::Tensor qcat, torch::Tensor normtab,
 void knn_graph(torch::Tensor coors, torch::Tensor nmask, torch::Tensor idx,
                torch::Tensor dist, torch::Tensor rel, torch::Tensor m,
                int64_t k, double radius, bool causal);
-void attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
-              torch::Tensor mask, torch::Tensor out,
-              int64_t n, int64_t heads, double scale, bool kv_one);
+void attn2_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+               torch::Tensor mask, torch::Tensor qf, torch::Tensor kf,
+               torch::Tensor out, torch::Tensor lse,
+               int64_t n, int64_t heads, double scale, bool kv_one,
+               int64_t jr, int64_t rot);
+void attn2_bwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+               torch::Tensor mask, torch::Tensor qf, torch::Tensor kf,
+               torch::Tensor out, torch::Tensor lse, torch::Tensor g,
+               torch::Tensor dq, torch::Tensor dk, torch::Tensor dv,
+               int64_t n, int64_t heads, double scale, bool kv_one,
+               int64_t jr, int64_t rot);
 void norm_se3_fwd(torch::Tensor t, torch::Tensor scale, torch::Tensor out, double eps);
 void norm_se3_bwd(torch::Tensor t, torch::Tensor scale, torch::Tensor dout,
                   torch::Tensor dt, torch::Tensor dscale, double eps);
@@ -392,7 +421,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("pairconv_bwd_dw", &pairconv_bwd_dw, "dW backward");
     m.def("pairconv_bwd_du", &pairconv_bwd_du, "dU backward");
     m.def("knn_graph", &knn_graph, "on-device kNN graph build");
-    m.def("attn_fwd", &attn_fwd, "fused neighbor attention forward");
+    m.def("attn2_fwd", &attn2_fwd,
+          "fused neighbor attention fwd (online softmax, in-kernel rotary)");
+    m.def("attn2_bwd", &attn2_bwd, "fused neighbor attention backward");
     m.def("norm_se3_fwd", &norm_se3_fwd, "fused NormSE3 forward");
     m.def("norm_se3_bwd", &norm_se3_bwd, "fused NormSE3 backward");
     m.def("sh_basis_fwd", &sh_basis_fwd,

@@ -23,6 +23,34 @@ def _first_tensor(features):
     return next(iter(features.values()))
 
 
+def _attn_kernel_ok(q, J, dm):
+    """Gate for the fused HIP attention kernel (csrc/attn2.hip): any J via
+    online-softmax tiles; DM bounded by the per-lane register chunking."""
+    from ..ops import fused as _fused
+    max_j = int(os.environ.get('SE3_ATTN_MAX_J', 100000))
+    return (q.is_cuda and J <= max_j and dm <= 448
+            and q.dtype in (torch.float32, torch.bfloat16)
+            and os.environ.get('SE3_EAGER_ATTN') != '1'
+            and _fused.ext_available())
+
+
+def _rope_in_kernel_ok(q, j_pre, qpe, kpe, use_null_kv, global_feats):
+    """True if the rotary embedding can be folded into the attention kernel
+    (q/k/v rotated in-registers) instead of materializing rotated copies.
+    The frequency tables must not need grad (rotary_rel_dist with
+    differentiable coordinates falls back to the eager path)."""
+    rot = qpe.shape[-1]
+    gj = 0
+    if global_feats is not None:   # fiber dict {'0': (b, gj, d, 1)}
+        gj = _first_tensor(global_feats).shape[1]
+    j_fin = j_pre + (1 if use_null_kv else 0) + gj
+    dm = q.shape[-2] * q.shape[-1]
+    return (_attn_kernel_ok(q, j_fin, dm)
+            and rot <= 64 and rot % 2 == 0 and rot <= dm
+            and kpe.shape[2] == j_pre
+            and not qpe.requires_grad and not kpe.requires_grad)
+
+
 class AttentionSE3(nn.Module):
     def __init__(self, fiber, dim_head=64, heads=8, attend_self=False,
                  edge_dim=None, fourier_encode_dist=False,
@@ -123,13 +151,23 @@ class AttentionSE3(nn.Module):
                 k = torch.cat((self_k, k), dim=3)
                 v = torch.cat((self_v, v), dim=3)
 
+            rope_q = rope_k = None
             if pos_emb is not None and degree == '0':
                 query_pos_emb, key_pos_emb = pos_emb
-                query_pos_emb = query_pos_emb[:, None, :, :, None]          # b 1 i d 1
-                key_pos_emb = key_pos_emb[:, None, :, :, :, None]           # b 1 i j d 1
-                q = apply_rotary_pos_emb(q, query_pos_emb)
-                k = apply_rotary_pos_emb(k, key_pos_emb)
-                v = apply_rotary_pos_emb(v, key_pos_emb)
+                if _rope_in_kernel_ok(q, k.shape[3], query_pos_emb, key_pos_emb,
+                                      self.use_null_kv, global_feats):
+                    # rotary folds into the attention kernel (q, k AND v are
+                    # rotated there, reference :488-494) — no eager copies
+                    rot_dim = query_pos_emb.shape[-1]
+                    rope_q = query_pos_emb.reshape(-1, rot_dim).float().contiguous()
+                    rope_k = key_pos_emb.reshape(-1, key_pos_emb.shape[2],
+                                                 rot_dim).float().contiguous()
+                else:
+                    query_pos_emb = query_pos_emb[:, None, :, :, None]      # b 1 i d 1
+                    key_pos_emb = key_pos_emb[:, None, :, :, :, None]       # b 1 i j d 1
+                    q = apply_rotary_pos_emb(q, query_pos_emb)
+                    k = apply_rotary_pos_emb(k, key_pos_emb)
+                    v = apply_rotary_pos_emb(v, key_pos_emb)
 
             if self.use_null_kv:
                 null_k, null_v = self.null_keys[degree], self.null_values[degree]
@@ -151,10 +189,7 @@ class AttentionSE3(nn.Module):
             bq, hq, nq, dq, mq = q.shape
             J = k.shape[3]
             from ..ops import fused as _fused
-            if (q.is_cuda and J <= 64 and dq * mq <= 448
-                    and q.dtype in (torch.float32, torch.bfloat16)
-                    and os.environ.get('SE3_EAGER_ATTN') != '1'
-                    and _fused.ext_available()):
+            if _attn_kernel_ok(q, J, dq * mq) or rope_q is not None:
                 mask_u8 = None
                 if neighbor_mask is not None:
                     pad = J - neighbor_mask.shape[-1]
@@ -164,7 +199,7 @@ class AttentionSE3(nn.Module):
                     q.reshape(bq * hq * nq, dq * mq).contiguous(),
                     k.to(q.dtype).reshape(bq * hq * nq, J, dq * mq).contiguous(),
                     v.to(q.dtype).reshape(bq * hq * nq, J, dq * mq).contiguous(),
-                    mask_u8, nq, hq, self.scale)
+                    mask_u8, nq, hq, self.scale, qf=rope_q, kf=rope_k)
                 out = out.view(bq, hq, nq, dq, mq).to(q.dtype)
             else:
                 sim = torch.einsum('bhidm,bhijdm->bhij', q, k) * self.scale
@@ -279,13 +314,21 @@ class OneHeadedKVAttentionSE3(nn.Module):
                 k = torch.cat((self_k.unsqueeze(2), k), dim=2)
                 v = torch.cat((self_v.unsqueeze(2), v), dim=2)
 
+            rope_q = rope_k = None
             if pos_emb is not None and degree == '0':
                 query_pos_emb, key_pos_emb = pos_emb
-                query_pos_emb = query_pos_emb[:, None, :, :, None]
-                key_pos_emb = key_pos_emb[..., None]
-                q = apply_rotary_pos_emb(q, query_pos_emb)
-                k = apply_rotary_pos_emb(k, key_pos_emb)
-                v = apply_rotary_pos_emb(v, key_pos_emb)
+                if _rope_in_kernel_ok(q, k.shape[2], query_pos_emb, key_pos_emb,
+                                      self.use_null_kv, global_feats):
+                    rot_dim = query_pos_emb.shape[-1]
+                    rope_q = query_pos_emb.reshape(-1, rot_dim).float().contiguous()
+                    rope_k = key_pos_emb.reshape(-1, key_pos_emb.shape[2],
+                                                 rot_dim).float().contiguous()
+                else:
+                    query_pos_emb = query_pos_emb[:, None, :, :, None]
+                    key_pos_emb = key_pos_emb[..., None]
+                    q = apply_rotary_pos_emb(q, query_pos_emb)
+                    k = apply_rotary_pos_emb(k, key_pos_emb)
+                    v = apply_rotary_pos_emb(v, key_pos_emb)
 
             if self.use_null_kv:
                 null_k, null_v = self.null_keys[degree], self.null_values[degree]
@@ -304,10 +347,7 @@ class OneHeadedKVAttentionSE3(nn.Module):
             bq, hq, nq, dq, mq = q.shape
             J = k.shape[2]
             from ..ops import fused as _fused
-            if (q.is_cuda and J <= 64 and dq * mq <= 448
-                    and q.dtype in (torch.float32, torch.bfloat16)
-                    and os.environ.get('SE3_EAGER_ATTN') != '1'
-                    and _fused.ext_available()):
+            if _attn_kernel_ok(q, J, dq * mq) or rope_q is not None:
                 mask_u8 = None
                 if neighbor_mask is not None:
                     pad = J - neighbor_mask.shape[-1]
@@ -317,7 +357,7 @@ class OneHeadedKVAttentionSE3(nn.Module):
                     q.reshape(bq * hq * nq, dq * mq).contiguous(),
                     k.to(q.dtype).reshape(bq * nq, J, dq * mq).contiguous(),
                     v.to(q.dtype).reshape(bq * nq, J, dq * mq).contiguous(),
-                    mask_u8, nq, hq, self.scale, True)
+                    mask_u8, nq, hq, self.scale, True, qf=rope_q, kf=rope_k)
                 out = out.view(bq, hq, nq, dq, mq).to(q.dtype)
             else:
                 sim = torch.einsum('bhidm,bijdm->bhij', q, k) * self.scale

@@ -223,67 +223,61 @@ def norm_se3(t, scale, eps):
 
 
 class _AttnFn(torch.autograd.Function):
-    """Fused neighbor attention (csrc/attn.hip): rows (b,h,i), keys J<=64.
-    Backward is the standard softmax-attention adjoint in library ops."""
+    """Fused neighbor attention v2 (csrc/attn2.hip): rows (b,h,i), keys in
+    64-wide online-softmax tiles (J unbounded), rotary embeddings applied to
+    q/k/v in-kernel on degree 0, HIP backward via the saved logsumexp.
+    qf/kf are the RAW rotary frequency tables (or None); gradients do not
+    flow into them (the wrapper gates on their requires_grad)."""
 
     @staticmethod
-    def forward(ctx, q, k, v, mask_u8, n, heads, scale, kv_one=False):
+    def forward(ctx, q, k, v, mask_u8, qf, kf, n, heads, scale, kv_one=False):
         ext = _load_ext()
         R, DM = q.shape
-        out = torch.empty(R, DM, dtype=torch.float32, device=q.device)
+        dev = q.device
+        out = torch.empty(R, DM, dtype=torch.float32, device=dev)
+        lse = torch.empty(R, dtype=torch.float32, device=dev)
         m = mask_u8 if mask_u8 is not None else \
-            torch.empty(0, dtype=torch.uint8, device=q.device)
-        ext.attn_fwd(q, k, v, m, out, n, heads, scale, kv_one)
-        ctx.save_for_backward(q, k, v,
-                              mask_u8 if mask_u8 is not None else None)
-        ctx.dims = (n, heads, scale, kv_one)
+            torch.empty(0, dtype=torch.uint8, device=dev)
+        qf_ = qf if qf is not None else torch.empty(0, device=dev)
+        kf_ = kf if kf is not None else torch.empty(0, device=dev)
+        jr = kf.shape[1] if kf is not None else 0
+        rot = (qf.shape[-1] if qf is not None
+               else (kf.shape[-1] if kf is not None else 0))
+        ext.attn2_fwd(q, k, v, m, qf_, kf_, out, lse,
+                      n, heads, scale, kv_one, jr, rot)
+        ctx.save_for_backward(q, k, v, mask_u8, qf, kf, out, lse)
+        ctx.dims = (n, heads, scale, kv_one, jr, rot)
         return out
 
     @staticmethod
     def backward(ctx, g):
-        q, k, v, mask_u8 = ctx.saved_tensors
-        n, heads, scale, kv_one = ctx.dims
-        Rkv, J, DM = k.shape
-        R = q.shape[0]
-        b = R // (heads * n)
-        qf, gf = q.float(), g.contiguous().float()
+        ext = _load_ext()
+        q, k, v, mask_u8, qf, kf, out, lse = ctx.saved_tensors
+        n, heads, scale, kv_one, jr, rot = ctx.dims
+        R, DM = q.shape
+        dev = q.device
+        m = mask_u8 if mask_u8 is not None else \
+            torch.empty(0, dtype=torch.uint8, device=dev)
+        qf_ = qf if qf is not None else torch.empty(0, device=dev)
+        kf_ = kf if kf is not None else torch.empty(0, device=dev)
+        dq = torch.empty(R, DM, dtype=torch.float32, device=dev)
         if kv_one:
-            kf = k.float().view(b, 1, n, J, DM)
-            vf = v.float().view(b, 1, n, J, DM)
-            q4 = qf.view(b, heads, n, DM)
-            g4 = gf.view(b, heads, n, DM)
-            sim = torch.einsum('bhnd,bxnjd->bhnj', q4, kf) * scale
-            if mask_u8 is not None:
-                mrows = mask_u8.view(b, 1, n, J).bool()
-                sim = sim.masked_fill(~mrows, -torch.finfo(sim.dtype).max)
-            attn = sim.softmax(dim=-1)
-            dv = torch.einsum('bhnj,bhnd->bnjd', attn, g4)
-            dattn = torch.einsum('bhnd,bxnjd->bhnj', g4, vf)
-            dsim = attn * (dattn - (attn * dattn).sum(-1, keepdim=True))
-            dq = torch.einsum('bhnj,bxnjd->bhnd', dsim, kf) * scale
-            dk = torch.einsum('bhnj,bhnd->bnjd', dsim, q4) * scale
-            return (dq.reshape(R, DM).to(q.dtype),
-                    dk.reshape(Rkv, J, DM).to(k.dtype),
-                    dv.reshape(Rkv, J, DM).to(v.dtype),
-                    None, None, None, None, None)
-        kf, vf = k.float(), v.float()
-        sim = torch.einsum('rd,rjd->rj', qf, kf) * scale
-        if mask_u8 is not None:
-            mrows = mask_u8.view(b, 1, n, J).expand(b, heads, n, J) \
-                .reshape(R, J).bool()
-            sim = sim.masked_fill(~mrows, -torch.finfo(sim.dtype).max)
-        attn = sim.softmax(dim=-1)
-        dv = torch.einsum('rj,rd->rjd', attn, gf)
-        dattn = torch.einsum('rd,rjd->rj', gf, vf)
-        dsim = attn * (dattn - (attn * dattn).sum(-1, keepdim=True))
-        dq = torch.einsum('rj,rjd->rd', dsim, kf) * scale
-        dk = torch.einsum('rj,rd->rjd', dsim, qf) * scale
+            # heads share KV rows: the kernel accumulates atomically
+            dk = torch.zeros(k.shape, dtype=torch.float32, device=dev)
+            dv = torch.zeros(v.shape, dtype=torch.float32, device=dev)
+        else:
+            dk = torch.empty(k.shape, dtype=torch.float32, device=dev)
+            dv = torch.empty(v.shape, dtype=torch.float32, device=dev)
+        ext.attn2_bwd(q, k, v, m, qf_, kf_, out, lse,
+                      g.contiguous().float(), dq, dk, dv,
+                      n, heads, scale, kv_one, jr, rot)
         return (dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype),
-                None, None, None, None, None)
+                None, None, None, None, None, None, None)
 
 
-def fused_attention(q, k, v, mask_u8, n, heads, scale, kv_one=False):
-    return _AttnFn.apply(q, k, v, mask_u8, n, heads, scale, kv_one)
+def fused_attention(q, k, v, mask_u8, n, heads, scale, kv_one=False,
+                    qf=None, kf=None):
+    return _AttnFn.apply(q, k, v, mask_u8, qf, kf, n, heads, scale, kv_one)
 
 
 RADIAL_TRUNK_DIMS = (1, 2, 3, 9, 17)   # instantiated in csrc/radial.hip

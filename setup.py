@@ -29,7 +29,7 @@ setup(
                      'se3_transformer_amd/csrc/radial.hip',
                      'se3_transformer_amd/csrc/sh_basis.hip',
                      'se3_transformer_amd/csrc/norm_se3.hip',
-                     'se3_transformer_amd/csrc/attn.hip',
+                     'se3_transformer_amd/csrc/attn2.hip',
                      'se3_transformer_amd/csrc/knn.hip'],
             extra_compile_args={
                 'cxx': ['-O3'],
