@@ -41,8 +41,14 @@ import torch  # noqa: E402
 
 
 def stage_reference(tmp='/tmp/ref_baseline'):
-    """Copy the reference package + shims into an importable temp tree."""
+    """Copy the reference package + shims into an importable temp tree.
+
+    On GPU boxes /root/reference is not mounted; the caller ships it via
+    the git-ignored .ref_stage/ snapshot directory instead (kept out of
+    the repo history — measurement input, not framework code)."""
     ref_src = '/root/reference/se3_transformer_pytorch'
+    if not os.path.isdir(ref_src):
+        ref_src = os.path.join(REPO, '.ref_stage', 'se3_transformer_pytorch')
     if not os.path.isdir(ref_src):
         raise SystemExit('reference not present on this box: ' + ref_src)
     pkg = os.path.join(tmp, 'se3_transformer_pytorch')
