@@ -379,7 +379,8 @@ void pairconv_fwd(torch::Tensor H, torch::Tensor W, torch::Tensor Ut,
 
 void sh_basis_fwd(torch::Tensor rel, torch::Tensor qcat, torch::Tensor normtab,
                   torch::Tensor meta, torch::Tensor out, int64_t L);
-void knn_graph(torch::Tensor coors, torch::Tensor nmask, torch::Tensor idx,
+void knn_graph(torch::Tensor coors, torch::Tensor nmask, torch::Tensor allow,
+               torch::Tensor sparse, torch::Tensor idx,
                torch::Tensor dist, torch::Tensor rel, torch::Tensor m,
                int64_t k, double radius, bool causal);
 void attn2_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,

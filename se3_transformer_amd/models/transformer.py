@@ -294,38 +294,60 @@ class SE3Transformer(nn.Module):
             sparse_neighbor_mask = torch.zeros_like(adj_values).scatter_(-1, indices, values) > 0.5
 
         # on-device kNN kernel path (csrc/knn.hip): replaces the dense
-        # (b,n,n) rel-geometry build + masked topk + gathers for the common
-        # configuration; selection semantics identical to the eager branch.
+        # (b,n,n) rel-geometry build + masked topk + gathers; selection
+        # semantics identical to the eager branch, including neighbor_mask
+        # exclusion, sparse-adjacency priority and causal masking. (The
+        # eager path's advisory print when neighbor_mask exceeds
+        # `neighbors` is skipped here.)
         from ..ops import fused as _fusedmod
-        k_eff = int(min(neighbors, n - 1))
+        k_total = int(min(neighbors + num_sparse_neighbors, n - 1))  # neighbors may be inf
         use_knn = (coors.is_cuda and coors.dtype == torch.float32
                    and not self.differentiable_coors
-                   and sparse_neighbor_mask is None
-                   and not exists(neighbor_mask)
-                   and neighbors > 0 and 1 <= k_eff <= 16
-                   and self.adj_emb is None
+                   and 1 <= k_total <= 64
                    and os.environ.get('SE3_EAGER_KNN') != '1'
                    and _fusedmod.ext_available())
 
         if use_knn:
+            k_eff = k_total
             idx = torch.empty(b, n, k_eff, dtype=torch.int64, device=device)
             dist = torch.empty(b, n, k_eff, device=device)
             relp = torch.empty(b, n, k_eff, 3, device=device)
             nm = torch.empty(b, n, k_eff, dtype=torch.uint8, device=device)
+            empty_u8 = torch.empty(0, dtype=torch.uint8, device=device)
             node_mask_u8 = mask.to(torch.uint8).contiguous() if exists(mask) \
-                else torch.empty(0, dtype=torch.uint8, device=device)
-            _fusedmod._EXT.knn_graph(coors.contiguous(), node_mask_u8, idx,
+                else empty_u8
+            allow_u8 = neighbor_mask.to(torch.uint8).contiguous() \
+                if exists(neighbor_mask) else empty_u8
+            sparse_u8 = empty_u8
+            if exists(sparse_neighbor_mask):
+                # off-diagonal (b,n,n-1) -> full-matrix (b,n,n) layout
+                full = torch.zeros(b, n, n, dtype=torch.bool, device=device)
+                full.scatter_(2, off_diag.unsqueeze(0).expand(b, n, n - 1),
+                              sparse_neighbor_mask)
+                sparse_u8 = full.to(torch.uint8).contiguous()
+            radius = float(valid_radius) if neighbors > 0 else 0.0
+            _fusedmod._EXT.knn_graph(coors.contiguous(), node_mask_u8,
+                                     allow_u8, sparse_u8, idx,
                                      dist, relp, nm, k_eff,
-                                     float(valid_radius), bool(self.causal))
+                                     radius, bool(self.causal))
             neighbor_indices = idx
             neighbor_rel_dist = dist
             neighbor_rel_pos = relp
             neighbor_mask = nm.bool()
 
+            if exists(edges) and exists(self.edge_emb):
+                edges = self.edge_emb(edges)
             if exists(edges):
-                if exists(self.edge_emb):
-                    edges = self.edge_emb(edges)
                 edges = batched_index_select(edges, neighbor_indices, dim=2)
+            if exists(self.adj_emb):
+                # adj_indices is off-diagonal (b,n,n-1): map the selected
+                # full-matrix j back to its off-diagonal column (j>i shifts)
+                iar = torch.arange(n, device=device).view(1, n, 1)
+                cols = neighbor_indices - (neighbor_indices > iar).long()
+                adj_e = self.adj_emb(batched_index_select(adj_indices, cols,
+                                                          dim=2))
+                edges = torch.cat((edges, adj_e), dim=-1) if exists(edges) \
+                    else adj_e
         else:
             # relative geometry (self excluded by indexing, not masked_select)
             indices = off_diag.unsqueeze(0).expand(b, n, n - 1)

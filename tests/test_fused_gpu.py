@@ -481,3 +481,55 @@ def test_attn2_rotary_in_kernel_vs_eager():
     out = model(feats, coors, mask, return_type=0)
     err = _rel_err(out, ref)
     assert err < 2e-3, f'rotary-in-kernel parity: {err}'
+
+
+@needs_gpu
+def test_knn_kernel_large_k_and_neighbor_mask():
+    """kNN kernel with k > 16 and a user neighbor_mask restricting
+    candidate selection (both round-2 extensions), vs the eager build."""
+    import os as _os
+    from se3_transformer_amd import SE3Transformer
+    torch.manual_seed(11)
+    model = SE3Transformer(dim=32, heads=2, dim_head=16, depth=1,
+                           num_degrees=2, num_neighbors=40,
+                           valid_radius=4.).to('cuda')
+    b, n = 2, 150
+    feats = torch.randn(b, n, 32, device='cuda')
+    coors = torch.randn(b, n, 3, device='cuda') * 1.5
+    mask = torch.ones(b, n, dtype=torch.bool, device='cuda')
+    nbr_allow = torch.rand(b, n, n, device='cuda') > 0.3
+    _os.environ['SE3_EAGER_KNN'] = '1'
+    try:
+        ref = model(feats, coors, mask, neighbor_mask=nbr_allow, return_type=0)
+    finally:
+        del _os.environ['SE3_EAGER_KNN']
+    out = model(feats, coors, mask, neighbor_mask=nbr_allow, return_type=0)
+    err = (out - ref).abs().max().item()
+    assert err < 1e-4, f'large-k/neighbor_mask knn mismatch: {err}'
+
+
+@needs_gpu
+def test_knn_kernel_sparse_adjacency():
+    """kNN kernel with attend_sparse_neighbors + adjacency embeddings
+    (qm9-style config) vs the eager build."""
+    import os as _os
+    from se3_transformer_amd import SE3Transformer
+    torch.manual_seed(12)
+    model = SE3Transformer(dim=32, heads=2, dim_head=16, depth=1,
+                           num_degrees=2, num_neighbors=4,
+                           attend_sparse_neighbors=True, num_adj_degrees=2,
+                           adj_dim=4, valid_radius=5.).to('cuda')
+    b, n = 2, 29
+    i = torch.arange(n, device='cuda')
+    adj_mat = ((i[:, None] - i[None, :]).abs() == 1)
+    feats = torch.randn(b, n, 32, device='cuda')
+    coors = torch.randn(b, n, 3, device='cuda')
+    mask = torch.ones(b, n, dtype=torch.bool, device='cuda')
+    _os.environ['SE3_EAGER_KNN'] = '1'
+    try:
+        ref = model(feats, coors, mask, adj_mat=adj_mat, return_type=0)
+    finally:
+        del _os.environ['SE3_EAGER_KNN']
+    out = model(feats, coors, mask, adj_mat=adj_mat, return_type=0)
+    err = (out - ref).abs().max().item()
+    assert err < 1e-4, f'sparse-adjacency knn mismatch: {err}'
