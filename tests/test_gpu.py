@@ -120,3 +120,49 @@ def test_differentiable_coors_cuda():
     out.pow(2).mean().backward()
     torch.cuda.synchronize()
     assert coors.grad is not None and torch.isfinite(coors.grad).all()
+
+
+@needs_gpu
+def test_equivariance_fused_bf16_autocast():
+    """Rotation equivariance of the FULL model with every fused kernel
+    active under autocast-bf16. The tolerance band is set
+    by bf16 resolution (2^-8) accumulated across depth-2/16-degree-pair
+    compute: measured fp32-fused error is ~1e-4; bf16 adds ~1e-2-scale
+    rounding, so 6e-2 relative on the max element is the principled band."""
+    from se3_transformer_amd.ops.wigner import rot
+    torch.manual_seed(21)
+    model = SE3Transformer(dim=64, heads=4, dim_head=16, depth=2,
+                           num_degrees=3, num_neighbors=8,
+                           attend_self=True).to('cuda')
+    feats = torch.randn(1, 48, 64, device='cuda')
+    coors = torch.randn(1, 48, 3, device='cuda') * 1.5
+    mask = torch.ones(1, 48, dtype=torch.bool, device='cuda')
+    R = rot(23., 117., 195.).float().to('cuda')
+    with torch.autocast(device_type='cuda', dtype=torch.bfloat16):
+        out1 = model(feats, coors @ R, mask, return_type=1)
+        out2 = model(feats, coors, mask, return_type=1)
+    out2 = out2.float() @ R
+    denom = out2.abs().max().clamp(min=1e-6)
+    err = ((out1.float() - out2).abs().max() / denom).item()
+    assert err < 6e-2, f'bf16 fused equivariance: {err}'
+
+
+@needs_gpu
+def test_equivariance_num_degrees_4_f64_cuda():
+    """Degree-4 float64 equivariance ON DEVICE (1e-8), mirroring the CPU
+    test at tests/test_equivariance.py (reference strictest regime)."""
+    from se3_transformer_amd.ops.wigner import rot
+    from se3_transformer_amd.utils import torch_default_dtype
+    with torch_default_dtype(torch.float64):
+        torch.manual_seed(0)
+        model = SE3Transformer(dim=8, heads=2, dim_head=4, depth=1,
+                               attend_self=True, num_neighbors=4,
+                               num_degrees=4, output_degrees=2).to('cuda')
+        feats = torch.randn(1, 12, 8, device='cuda')
+        coors = torch.randn(1, 12, 3, device='cuda')
+        mask = torch.ones(1, 12, dtype=torch.bool, device='cuda')
+        R = rot(23., 117., 195.).to(torch.float64).to('cuda')
+        out1 = model(feats, coors @ R, mask, return_type=1)
+        out2 = model(feats, coors, mask, return_type=1) @ R
+        diff = (out1 - out2).abs().max().item()
+        assert diff < 1e-8, f'degree-4 f64 equivariance on GPU: {diff}'
