@@ -45,9 +45,12 @@ pairconv_bwd_dh_kernel(const __bf16* __restrict__ Gt,  // (mo, O, E) bf16
                        float* __restrict__ dH,         // (E, 128) f32 (zeroed)
                        int E, int mo, int miF, int nsplit, int nmemb, int coh) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
+    // u/g tiles are [row][e][o PADDED to 8]: the dR build then reads ONE
+    // 16B vector per (row, e) and contracts with v_dot2_f32_bf16 — no
+    // scalar LDS reads, no bf16->f32 conversion ops.
     __bf16* dr_lds = reinterpret_cast<__bf16*>(smem);                 // [64e][256n] 32 KiB
-    __bf16* u_lds = reinterpret_cast<__bf16*>(smem + 32768);          // [32][O][64]
-    __bf16* g_lds = reinterpret_cast<__bf16*>(smem + 32768 + 32 * O * 64 * 2); // [8mo][O][64e]
+    __bf16* u_lds = reinterpret_cast<__bf16*>(smem + 32768);          // [32][64][8] 32 KiB
+    __bf16* g_lds = reinterpret_cast<__bf16*>(smem + 65536);          // [8][64][8] 8 KiB
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -110,16 +113,22 @@ pairconv_bwd_dh_kernel(const __bf16* __restrict__ Gt,  // (mo, O, E) bf16
     };
     load_u(0);
     load_g(mb_lo);
+    // zero the o-pad lanes once (staging never overwrites o in [O, 8))
+    for (int i = tid; i < (32 + 8) * 64; i += NT)
+        *reinterpret_cast<bf16x8*>(u_lds + (size_t)i * 8) = bf16x8(0);
 
     for (int cb = 0; cb < nuc; ++cb) {
-        // commit the staged u chunk [32][O][64] once per urow-chunk
+        // commit the staged u chunk, scattered to [urow][e][o(pad 8)]
         __syncthreads();
 #pragma unroll
         for (int t = 0; t < UUD; ++t) {
             int i = tid + t * NT;
             if (i < UTOTd) {
                 int ro = i >> 3, eu = (i & 7) * 8;
-                *reinterpret_cast<bf16x8*>(u_lds + (size_t)ro * 64 + eu) = u_reg[t];
+                const int ur = ro / O, o = ro % O;
+#pragma unroll
+                for (int j = 0; j < 8; ++j)
+                    u_lds[((size_t)ur * 64 + eu + j) * 8 + o] = u_reg[t][j];
             }
         }
         for (int i = tid + UUD * NT; i < UTOTd; i += NT) {   // unstaged tail
@@ -128,45 +137,47 @@ pairconv_bwd_dh_kernel(const __bf16* __restrict__ Gt,  // (mo, O, E) bf16
             bf16x8 v;
             if (e0 + eu + 8 <= E) v = *reinterpret_cast<const bf16x8*>(src);
             else { for (int j = 0; j < 8; ++j) v[j] = (e0 + eu + j < E) ? src[j] : (__bf16)0.f; }
-            *reinterpret_cast<bf16x8*>(u_lds + (size_t)ro * 64 + eu) = v;
+            const int ur = ro / O, o = ro % O;
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+                u_lds[((size_t)ur * 64 + eu + j) * 8 + o] = v[j];
         }
         if (cb + 1 < nuc) load_u(cb + 1);
         for (int mb = mb_lo; mb < mb_hi; ++mb) {
-            // commit the staged g tile [8][O][64]
+            // commit the staged g tile, scattered to [m][e][o(pad 8)]
             if (tid < GTOT) {
                 int ro = tid >> 3, eu = (tid & 7) * 8;
-                *reinterpret_cast<bf16x8*>(g_lds + (size_t)ro * 64 + eu) = g_reg;
+                const int mr = ro / O, o = ro % O;
+#pragma unroll
+                for (int j = 0; j < 8; ++j)
+                    g_lds[((size_t)mr * 64 + eu + j) * 8 + o] = g_reg[j];
             }
             load_g(mb + 1 < mb_hi ? mb + 1 : mb_lo);   // next mb (or next cb's first)
             __syncthreads();
-            // cooperative dR tile: [64e][256n]; thread owns (e-pair, run of 8 n):
-            // b32 LDS reads over e-pairs + packed fma, two swizzled b128 writes.
-            for (int i = tid; i < (64 * 256) / 16; i += NT) {
-                int e = (i & 31) * 2, n8 = i >> 5;
+            // cooperative dR tile [64e][256n]: thread owns (e, run of 8 n);
+            // per n: one 16B g vector (shared over the run) dot one 16B u
+            // vector via 4 v_dot2_f32_bf16; one swizzled b128 write per run.
+            for (int i = tid; i < (64 * 256) / 8; i += NT) {
+                int e = i & 63, n8 = i >> 6;
                 int m = n8 >> 2;                      // (n8*8)>>5
-                f32x2 gv[O];
-#pragma unroll
-                for (int o = 0; o < O; ++o)
-                    gv[o] = b2f2(*reinterpret_cast<const bf16x2*>(g_lds + (m * O + o) * 64 + e));
-                __bf16 v0[8], v1[8];
+                bf16x8 g8 = *reinterpret_cast<const bf16x8*>(
+                    g_lds + ((size_t)m * 64 + e) * 8);
+                const bf16x2* g2 = reinterpret_cast<const bf16x2*>(&g8);
+                __bf16 v0[8];
 #pragma unroll
                 for (int j = 0; j < 8; ++j) {
                     int c = (n8 * 8 + j) & 31;
-                    f32x2 acc2 = {0.f, 0.f};
+                    bf16x8 u8 = *reinterpret_cast<const bf16x8*>(
+                        u_lds + ((size_t)c * 64 + e) * 8);
+                    const bf16x2* u2 = reinterpret_cast<const bf16x2*>(&u8);
+                    float acc = 0.f;
 #pragma unroll
-                    for (int o = 0; o < O; ++o) {
-                        f32x2 uv = b2f2(*reinterpret_cast<const bf16x2*>(
-                            u_lds + (c * O + o) * 64 + e));
-                        acc2[0] = fmaf(gv[o][0], uv[0], acc2[0]);
-                        acc2[1] = fmaf(gv[o][1], uv[1], acc2[1]);
-                    }
-                    v0[j] = (__bf16)acc2[0];
-                    v1[j] = (__bf16)acc2[1];
+                    for (int p = 0; p < 4; ++p)
+                        acc = __builtin_amdgcn_fdot2_f32_bf16(g2[p], u2[p], acc, false);
+                    v0[j] = (__bf16)acc;
                 }
                 *reinterpret_cast<bf16x8*>(reinterpret_cast<char*>(dr_lds)
                     + e * 512 + (((n8 ^ (e & 15)) << 4))) = *reinterpret_cast<bf16x8*>(v0);
-                *reinterpret_cast<bf16x8*>(reinterpret_cast<char*>(dr_lds)
-                    + (e + 1) * 512 + (((n8 ^ ((e + 1) & 15)) << 4))) = *reinterpret_cast<bf16x8*>(v1);
             }
             __syncthreads();
             // MFMA: dH_tile += dR(64e x 256n) @ W(256n x 128k)
@@ -219,10 +230,12 @@ pairconv_bwd_dw_kernel(const __bf16* __restrict__ Gt,  // (mo, O, E)
                        float* __restrict__ dW,         // (mo*miF, 128) f32
                        int E, int mo, int miF) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
+    // u/g tiles are [row][e][o PADDED to 8] so the dR build reads one 16B
+    // vector per (row, e) and contracts with v_dot2_f32_bf16 (see dh)
     __bf16* dr_lds = reinterpret_cast<__bf16*>(smem);                  // [128n][32e] 8 KiB
     __bf16* h_lds = reinterpret_cast<__bf16*>(smem + 8192);            // [128k][32e] 8 KiB
-    __bf16* u_lds = reinterpret_cast<__bf16*>(smem + 16384);           // [32][O][32]
-    __bf16* g_lds = reinterpret_cast<__bf16*>(smem + 16384 + 32 * O * 32 * 2); // [4][O][32]
+    __bf16* u_lds = reinterpret_cast<__bf16*>(smem + 16384);           // [32][32][8] 16 KiB
+    __bf16* g_lds = reinterpret_cast<__bf16*>(smem + 32768);           // [4][32][8] 2 KiB
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -282,20 +295,29 @@ pairconv_bwd_dw_kernel(const __bf16* __restrict__ Gt,  // (mo, O, E)
     };
 
     load_chunk(0);
+    // zero the o-pad lanes once (staging never overwrites o in [O, 8))
+    for (int i = tid; i < (32 + 4) * 32; i += NT)
+        *reinterpret_cast<bf16x8*>(u_lds + (size_t)i * 8) = bf16x8(0);
     for (int ec = 0; ec < nec; ++ec) {
         __syncthreads();   // previous MFMA done reading the LDS images
-        // write the staged registers for chunk ec
+        // write the staged registers for chunk ec, scattered to o-padded form
 #pragma unroll
         for (int t = 0; t < UU; ++t) {
             int i = tid + t * NT;
             if (i < (32 * O * 32) / 8) {
                 int ro = i >> 2, eu = (i & 3) * 8;
-                *reinterpret_cast<bf16x8*>(u_lds + (size_t)ro * 32 + eu) = u_reg[t];
+                const int ur = ro / O, o = ro % O;
+#pragma unroll
+                for (int j = 0; j < 8; ++j)
+                    u_lds[((size_t)ur * 32 + eu + j) * 8 + o] = u_reg[t][j];
             }
         }
         if (tid < (4 * O * 32) / 8) {
             int ro = tid >> 2, eu = (tid & 3) * 8;
-            *reinterpret_cast<bf16x8*>(g_lds + (size_t)ro * 32 + eu) = g_reg[0];
+            const int mr = ro / O, o = ro % O;
+#pragma unroll
+            for (int j = 0; j < 8; ++j)
+                g_lds[((size_t)mr * 32 + eu + j) * 8 + o] = g_reg[0][j];
         }
         {
             int k = tid >> 2, eu = (tid & 3) * 8;
@@ -303,20 +325,22 @@ pairconv_bwd_dw_kernel(const __bf16* __restrict__ Gt,  // (mo, O, E)
         }
         if (ec + 1 < nec) load_chunk(ec + 1);   // issue next loads early
         __syncthreads();
-        // cooperative dR^T tile [128n][32e] (e-pairs, packed)
-        for (int i = tid; i < (128 * 32) / 2; i += NT) {
-            int e = (i & 15) * 2, n = i >> 4;
+        // cooperative dR^T tile [128n][32e]: one 16B g + one 16B u vector
+        // per (n, e), contracted over o with 4 v_dot2_f32_bf16
+        for (int i = tid; i < 128 * 32; i += NT) {
+            int e = i & 31, n = i >> 5;
             int m = n >> 5, c = n & 31;
-            f32x2 acc2 = {0.f, 0.f};
+            bf16x8 g8 = *reinterpret_cast<const bf16x8*>(
+                g_lds + ((size_t)m * 32 + e) * 8);
+            bf16x8 u8 = *reinterpret_cast<const bf16x8*>(
+                u_lds + ((size_t)c * 32 + e) * 8);
+            const bf16x2* g2 = reinterpret_cast<const bf16x2*>(&g8);
+            const bf16x2* u2 = reinterpret_cast<const bf16x2*>(&u8);
+            float a2 = 0.f;
 #pragma unroll
-            for (int o = 0; o < O; ++o) {
-                f32x2 gv = b2f2(*reinterpret_cast<const bf16x2*>(g_lds + (m * O + o) * 32 + e));
-                f32x2 uv = b2f2(*reinterpret_cast<const bf16x2*>(u_lds + (c * O + o) * 32 + e));
-                acc2[0] = fmaf(gv[0], uv[0], acc2[0]);
-                acc2[1] = fmaf(gv[1], uv[1], acc2[1]);
-            }
-            *reinterpret_cast<bf16x2*>(dr_lds + (size_t)n * 32 + e) =
-                bf16x2{(__bf16)acc2[0], (__bf16)acc2[1]};
+            for (int p = 0; p < 4; ++p)
+                a2 = __builtin_amdgcn_fdot2_f32_bf16(g2[p], u2[p], a2, false);
+            dr_lds[(size_t)n * 32 + e] = (__bf16)a2;
         }
         __syncthreads();
         // MFMA: dW_tile += dR^T(128n x 32e) @ H(32e x 128k)
@@ -530,7 +554,7 @@ void pairconv_bwd_dh(torch::Tensor Gt, torch::Tensor Ut, torch::Tensor Wt,
     int panels = ((eblk + PE - 1) / PE) * ((nsplit + PS - 1) / PS);
     dim3 grid((long)panels * PS * PE);
     DISPATCH_O(O, {
-        size_t lds = 32768 + (size_t)32 * kO * 64 * 2 + (size_t)8 * kO * 64 * 2;
+        size_t lds = 32768 + 32768 + 8192;   // dr | u[32][64][8] | g[8][64][8]
         hipLaunchKernelGGL(HIP_KERNEL_NAME(pairconv_bwd_dh_kernel<kO>), grid, dim3(NT), lds, stream,
                            reinterpret_cast<const __bf16*>(Gt.data_ptr()),
                            reinterpret_cast<const __bf16*>(Ut.data_ptr()),
@@ -550,7 +574,7 @@ void pairconv_bwd_dw(torch::Tensor Gt, torch::Tensor Ut, torch::Tensor Ht,
     auto stream = at::cuda::getCurrentHIPStream();
     dim3 grid((mo / 4) * (miF / 32));
     DISPATCH_O(O, {
-        size_t lds = 16384 + (size_t)32 * kO * 32 * 2 + (size_t)4 * kO * 32 * 2;
+        size_t lds = 16384 + 16384 + 2048;   // dr+h | u[32][32][8] | g[4][32][8]
         hipLaunchKernelGGL(HIP_KERNEL_NAME(pairconv_bwd_dw_kernel<kO>), grid, dim3(NT), lds, stream,
                            reinterpret_cast<const __bf16*>(Gt.data_ptr()),
                            reinterpret_cast<const __bf16*>(Ut.data_ptr()),
