@@ -230,12 +230,14 @@ pairconv_bwd_dw_kernel(const __bf16* __restrict__ Gt,  // (mo, O, E)
                        float* __restrict__ dW,         // (mo*miF, 128) f32
                        int E, int mo, int miF) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
-    // u/g tiles are [row][e][o PADDED to 8] so the dR build reads one 16B
-    // vector per (row, e) and contracts with v_dot2_f32_bf16 (see dh)
+    // NOTE: dw keeps the e-pair packed-fma dR build — the o-padded dot2
+    // layout (as in dh) was tried and measured SLOWER here (234 -> 187
+    // TF/s): dw stages per 32-EDGE chunk, so the scatter commit the padded
+    // layout needs runs every chunk and dominates.
     __bf16* dr_lds = reinterpret_cast<__bf16*>(smem);                  // [128n][32e] 8 KiB
     __bf16* h_lds = reinterpret_cast<__bf16*>(smem + 8192);            // [128k][32e] 8 KiB
-    __bf16* u_lds = reinterpret_cast<__bf16*>(smem + 16384);           // [32][32][8] 16 KiB
-    __bf16* g_lds = reinterpret_cast<__bf16*>(smem + 32768);           // [4][32][8] 2 KiB
+    __bf16* u_lds = reinterpret_cast<__bf16*>(smem + 16384);           // [32][O][32]
+    __bf16* g_lds = reinterpret_cast<__bf16*>(smem + 16384 + 32 * O * 32 * 2); // [4][O][32]
 
     const int tid = threadIdx.x;
     const int lane = tid & 63;
@@ -295,29 +297,20 @@ pairconv_bwd_dw_kernel(const __bf16* __restrict__ Gt,  // (mo, O, E)
     };
 
     load_chunk(0);
-    // zero the o-pad lanes once (staging never overwrites o in [O, 8))
-    for (int i = tid; i < (32 + 4) * 32; i += NT)
-        *reinterpret_cast<bf16x8*>(u_lds + (size_t)i * 8) = bf16x8(0);
     for (int ec = 0; ec < nec; ++ec) {
         __syncthreads();   // previous MFMA done reading the LDS images
-        // write the staged registers for chunk ec, scattered to o-padded form
+        // write the staged registers for chunk ec
 #pragma unroll
         for (int t = 0; t < UU; ++t) {
             int i = tid + t * NT;
             if (i < (32 * O * 32) / 8) {
                 int ro = i >> 2, eu = (i & 3) * 8;
-                const int ur = ro / O, o = ro % O;
-#pragma unroll
-                for (int j = 0; j < 8; ++j)
-                    u_lds[((size_t)ur * 32 + eu + j) * 8 + o] = u_reg[t][j];
+                *reinterpret_cast<bf16x8*>(u_lds + (size_t)ro * 32 + eu) = u_reg[t];
             }
         }
         if (tid < (4 * O * 32) / 8) {
             int ro = tid >> 2, eu = (tid & 3) * 8;
-            const int mr = ro / O, o = ro % O;
-#pragma unroll
-            for (int j = 0; j < 8; ++j)
-                g_lds[((size_t)mr * 32 + eu + j) * 8 + o] = g_reg[0][j];
+            *reinterpret_cast<bf16x8*>(g_lds + (size_t)ro * 32 + eu) = g_reg[0];
         }
         {
             int k = tid >> 2, eu = (tid & 3) * 8;
@@ -325,22 +318,20 @@ pairconv_bwd_dw_kernel(const __bf16* __restrict__ Gt,  // (mo, O, E)
         }
         if (ec + 1 < nec) load_chunk(ec + 1);   // issue next loads early
         __syncthreads();
-        // cooperative dR^T tile [128n][32e]: one 16B g + one 16B u vector
-        // per (n, e), contracted over o with 4 v_dot2_f32_bf16
-        for (int i = tid; i < 128 * 32; i += NT) {
-            int e = i & 31, n = i >> 5;
+        // cooperative dR^T tile [128n][32e] (e-pairs, packed)
+        for (int i = tid; i < (128 * 32) / 2; i += NT) {
+            int e = (i & 15) * 2, n = i >> 4;
             int m = n >> 5, c = n & 31;
-            bf16x8 g8 = *reinterpret_cast<const bf16x8*>(
-                g_lds + ((size_t)m * 32 + e) * 8);
-            bf16x8 u8 = *reinterpret_cast<const bf16x8*>(
-                u_lds + ((size_t)c * 32 + e) * 8);
-            const bf16x2* g2 = reinterpret_cast<const bf16x2*>(&g8);
-            const bf16x2* u2 = reinterpret_cast<const bf16x2*>(&u8);
-            float a2 = 0.f;
+            f32x2 acc2 = {0.f, 0.f};
 #pragma unroll
-            for (int p = 0; p < 4; ++p)
-                a2 = __builtin_amdgcn_fdot2_f32_bf16(g2[p], u2[p], a2, false);
-            dr_lds[(size_t)n * 32 + e] = (__bf16)a2;
+            for (int o = 0; o < O; ++o) {
+                f32x2 gv = b2f2(*reinterpret_cast<const bf16x2*>(g_lds + (m * O + o) * 32 + e));
+                f32x2 uv = b2f2(*reinterpret_cast<const bf16x2*>(u_lds + (c * O + o) * 32 + e));
+                acc2[0] = fmaf(gv[0], uv[0], acc2[0]);
+                acc2[1] = fmaf(gv[1], uv[1], acc2[1]);
+            }
+            *reinterpret_cast<bf16x2*>(dr_lds + (size_t)n * 32 + e) =
+                bf16x2{(__bf16)acc2[0], (__bf16)acc2[1]};
         }
         __syncthreads();
         // MFMA: dW_tile += dR^T(128n x 32e) @ H(32e x 128k)
@@ -574,7 +565,7 @@ void pairconv_bwd_dw(torch::Tensor Gt, torch::Tensor Ut, torch::Tensor Ht,
     auto stream = at::cuda::getCurrentHIPStream();
     dim3 grid((mo / 4) * (miF / 32));
     DISPATCH_O(O, {
-        size_t lds = 16384 + 16384 + 2048;   // dr+h | u[32][32][8] | g[4][32][8]
+        size_t lds = 16384 + (size_t)32 * kO * 32 * 2 + (size_t)4 * kO * 32 * 2;
         hipLaunchKernelGGL(HIP_KERNEL_NAME(pairconv_bwd_dw_kernel<kO>), grid, dim3(NT), lds, stream,
                            reinterpret_cast<const __bf16*>(Gt.data_ptr()),
                            reinterpret_cast<const __bf16*>(Ut.data_ptr()),

@@ -132,7 +132,7 @@ def test_equivariance_fused_bf16_autocast():
     from se3_transformer_amd.ops.wigner import rot
     torch.manual_seed(21)
     model = SE3Transformer(dim=64, heads=4, dim_head=16, depth=2,
-                           num_degrees=3, num_neighbors=8,
+                           num_degrees=3, num_neighbors=8, output_degrees=2,
                            attend_self=True).to('cuda')
     feats = torch.randn(1, 48, 64, device='cuda')
     coors = torch.randn(1, 48, 3, device='cuda') * 1.5
