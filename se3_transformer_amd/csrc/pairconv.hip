@@ -404,6 +404,10 @@ void pairconv_bwd_dw(torch::Tensor G, torch::Tensor Ut, torch::Tensor Ht,
 void pairconv_bwd_du(torch::Tensor H, torch::Tensor W, torch::Tensor bias,
                      torch::Tensor G, torch::Tensor dU, int64_t mo_);
 void pack_w_both(torch::Tensor W, torch::Tensor Pf, torch::Tensor Pdh, int64_t mo_);
+void ubuild_fwd(torch::Tensor B, torch::Tensor X, torch::Tensor Ut,
+                int64_t O, int64_t I, int64_t F);
+void ubuild_bwd_dx(torch::Tensor B, torch::Tensor dU, torch::Tensor dX,
+                   int64_t O, int64_t I, int64_t F);
 void radial_trunk_fwd(torch::Tensor X, torch::Tensor W0, torch::Tensor p0,
                       torch::Tensor W3, torch::Tensor p3,
                       torch::Tensor H, torch::Tensor yh0, torch::Tensor yh3,
@@ -435,4 +439,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "fused radial trunk (Linear-LN-GELU x2) forward");
     m.def("radial_trunk_bwd", &radial_trunk_bwd,
           "fused radial trunk backward");
+    m.def("ubuild_fwd", &ubuild_fwd,
+          "basis x features precontraction (e-contiguous out)");
+    m.def("ubuild_bwd_dx", &ubuild_bwd_dx, "ubuild dX backward");
 }

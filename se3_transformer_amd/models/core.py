@@ -228,7 +228,7 @@ class PairwiseConv(nn.Module):
             e_total *= s
 
         ef = edge_feats.reshape(e_total, edge_feats.shape[-1])
-        bx = b.reshape(e_total, O, I, F_).to(x_gathered.dtype)
+        braw = b.reshape(e_total, O, I, F_)
         xg = x_gathered.reshape(e_total, mi, I)
 
         # MI355X fused HIP path: bf16 compute on CUDA(ROCm) devices
@@ -241,13 +241,20 @@ class PairwiseConv(nn.Module):
             _fused.require_ext()
             h = self.rp.hidden(ef)                           # (E, 128)
             # u_t[(mi,f), o, e] = sum_i B[e,o,i,f] x[e,mi,i]
-            u_t = torch.einsum('eoif,eci->cfoe', bx, xg).reshape(mi * F_, O, e_total)
+            if (_fused.ubuild_ok(braw, mi, O, I, F_)
+                    and os.environ.get('SE3_EAGER_UBUILD') != '1'):
+                u_t = _fused.ubuild(xg, braw.contiguous(), O, I, F_)
+            else:
+                u_t = torch.einsum('eoif,eci->cfoe',
+                                   braw.to(xg.dtype), xg) \
+                    .reshape(mi * F_, O, e_total)
             w6 = self.rp.net[6]
             out = _fused.fused_pairconv(h, w6.weight, w6.bias, u_t, mo)
             return out.view(*lead, mo, O).to(xg.dtype)
 
         # u: (E, mi*F, O), (mi-major, f-minor) matching R's (mo, mi*F) layout
-        u = torch.einsum('eoif,eci->ecfo', bx, xg).reshape(e_total, mi * F_, O)
+        u = torch.einsum('eoif,eci->ecfo', braw.to(xg.dtype), xg) \
+            .reshape(e_total, mi * F_, O)
 
         elem_size = xg.element_size()
         per_edge = mo * mi * F_ * elem_size

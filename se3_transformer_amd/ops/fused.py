@@ -280,6 +280,43 @@ def fused_attention(q, k, v, mask_u8, n, heads, scale, kv_one=False,
     return _AttnFn.apply(q, k, v, mask_u8, qf, kf, n, heads, scale, kv_one)
 
 
+class _UBuildFn(torch.autograd.Function):
+    """Basis-feature precontraction (csrc/ubuild.hip):
+    Ut[(c f), o, e] = sum_i B[e,o,i,f] x[e,c,i], emitted directly in the
+    e-contiguous layout the pairconv kernels consume. dB is not produced
+    (the caller gates on basis.requires_grad)."""
+
+    @staticmethod
+    def forward(ctx, x, B, O, I, F):
+        ext = _load_ext()
+        E, C, _ = x.shape
+        Ut = torch.empty(C * F, O, E, dtype=torch.bfloat16, device=x.device)
+        ext.ubuild_fwd(B, x.contiguous(), Ut, O, I, F)
+        ctx.save_for_backward(B)
+        ctx.meta = (x.dtype, x.shape, O, I, F)
+        return Ut
+
+    @staticmethod
+    def backward(ctx, dUt):
+        ext = _load_ext()
+        (B,) = ctx.saved_tensors
+        dtype, (E, C, I_), O, I, F = ctx.meta
+        dX = torch.empty(E, C, I, dtype=dtype, device=B.device)
+        ext.ubuild_bwd_dx(B, dUt.contiguous().float(), dX, O, I, F)
+        return dX, None, None, None, None
+
+
+def ubuild_ok(B, C, O, I, F) -> bool:
+    ext = _load_ext()
+    return (ext is not None and hasattr(ext, 'ubuild_fwd')
+            and B.dtype == torch.float32 and not B.requires_grad
+            and C % 32 == 0 and O * I * F <= 343)
+
+
+def ubuild(x, B, O, I, F):
+    return _UBuildFn.apply(x, B, O, I, F)
+
+
 RADIAL_TRUNK_DIMS = (1, 2, 3, 9, 17)   # instantiated in csrc/radial.hip
 
 
