@@ -76,6 +76,13 @@ def main():
             print(f'fwd UU={uu} MB2={mb2} ({di},{do}): {ms:8.3f} ms  '
                   f'{gemm_fl/ms/1e9:7.1f} TF/s (gemm)')
         del os.environ['SE3_FWD_UU']
+        for mb2 in ('1', '2'):
+            os.environ['SE3_FWD_WP'] = '1'
+            os.environ['SE3_FWD_MB2'] = mb2
+            ms = timeit(lambda: _C.pairconv_fwd(H, P, Ut, out, mo))
+            print(f'fwd WP=1 MB2={mb2} ({di},{do}): {ms:8.3f} ms  '
+                  f'{gemm_fl/ms/1e9:7.1f} TF/s (gemm)')
+        del os.environ['SE3_FWD_WP']
         del os.environ['SE3_FWD_MB2']
     if args.only:
         fn = {'fwd': lambda: _C.pairconv_fwd(H, P, Ut, out, mo),

@@ -56,7 +56,12 @@ __device__ __forceinline__ float bf16_to_f32(unsigned short u) {
 // while keeping 2-blocks/CU residency: the two sub-blocks run serially, so
 // the MFMA accumulator registers are reused, only the LDS `part` doubles
 // (O=7: 16K h + 28K u + 28K part = 72 KiB <= 80 KiB per block).
-template <int O, int UU, int MB2>   // UU = 16B u-units register-staged per thread
+// WP = W-fragment software pipeline: the next sub-block's 16 A-fragments
+// are prefetched into registers during the current sub-block's epilogue
+// (where the MFMA accumulators are dead), so each MFMA phase starts with
+// its operands already in flight instead of paying a fresh L2/HBM round
+// trip (T14 applied to the W stream).
+template <int O, int UU, int MB2, int WP>   // UU = 16B u-units register-staged per thread
 __global__ void __launch_bounds__(NTHREADS, 4)   // cap VGPR<=128: 2 blocks/CU
 pairconv_fwd_kernel(const __bf16* __restrict__ H,
                     const __bf16* __restrict__ P,   // packed W: [mo/8][miF/32][wm4][mf4][kit4][lane64][8]
@@ -140,9 +145,21 @@ pairconv_fwd_kernel(const __bf16* __restrict__ H,
     };
 
     load_u(0);
-    __syncthreads();
 
     const int nchunks = miF / UCHUNK;
+    // W-pipeline state: the 16 A-fragments of the NEXT (chunk, sub) pair
+    auto pb_of = [&](int cc, int ss) {
+        return P + ((((size_t)(mb * MB2 + ss) * (miF / 32) + cc) * 4 + wm) * 4) * 4 * 64 * 8
+               + (size_t)lane * 8;
+    };
+    bf16x8 a_pre[WP ? 16 : 1];
+    if (WP) {
+#pragma unroll
+        for (int f = 0; f < 16; ++f)
+            a_pre[f] = *reinterpret_cast<const bf16x8*>(pb_of(0, 0) + (size_t)f * 512);
+    }
+    __syncthreads();
+
     for (int c = 0; c < nchunks; ++c) {
         // ---- commit the staged u chunk, scattered to [urow][e][o(pad 8)]
 #pragma unroll
@@ -191,15 +208,16 @@ pairconv_fwd_kernel(const __bf16* __restrict__ H,
 #pragma unroll
             for (int ef = 0; ef < 2; ++ef) acc[mf][ef] = f32x4(0.f);
 
-        const __bf16* pbase = P + ((((size_t)(mb * MB2 + sub) * (miF / 32) + c) * 4 + wm) * 4) * 4 * 64 * 8
-                              + (size_t)lane * 8;
+        const __bf16* pbase = pb_of(c, sub);
 #pragma unroll
         for (int kit = 0; kit < 4; ++kit) {
             const int k0 = kit * 32 + l4 * 8;
             bf16x8 a[4], b[2];
 #pragma unroll
             for (int mf = 0; mf < 4; ++mf) {
-                a[mf] = *reinterpret_cast<const bf16x8*>(pbase + ((size_t)mf * 4 + kit) * 64 * 8);
+                a[mf] = WP ? a_pre[mf * 4 + kit]
+                           : *reinterpret_cast<const bf16x8*>(
+                                 pbase + ((size_t)mf * 4 + kit) * 64 * 8);
             }
 #pragma unroll
             for (int ef = 0; ef < 2; ++ef) {
@@ -212,6 +230,18 @@ pairconv_fwd_kernel(const __bf16* __restrict__ H,
 #pragma unroll
                 for (int ef = 0; ef < 2; ++ef)
                     acc[mf][ef] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[mf], b[ef], acc[mf][ef], 0, 0, 0);
+        }
+        if (WP) {
+            // prefetch the NEXT sub-block/chunk's fragments now — their
+            // L2/HBM latency hides under the epilogue below (acc is the
+            // only other register consumer left, and it dies there)
+            int ns = sub + 1, nc2 = c;
+            if (ns == MB2) { ns = 0; ++nc2; }
+            if (nc2 == nchunks) { ns = sub; nc2 = c; }   // tail: benign refetch
+            const __bf16* pn = pb_of(nc2, ns);
+#pragma unroll
+            for (int f = 0; f < 16; ++f)
+                a_pre[f] = *reinterpret_cast<const bf16x8*>(pn + (size_t)f * 512);
         }
 
         // ---- epilogue: contract acc against u_lds into s[ef][moi][o-pairs].
@@ -321,29 +351,36 @@ static void launch_fwd(const torch::Tensor& H, const torch::Tensor& W,
     int mb2 = (mb_env ? atoi(mb_env) : 2);
     if (mb2 != 1 && mo % (BLK_MO * 2) != 0) mb2 = 1;
     if (mb2 != 1) mb2 = 2;
+    // WP: W-fragment software pipeline (prefetch next sub-block's A-frags
+    // under the epilogue). Only instantiated with UU=0.
+    const char* wp_env = getenv("SE3_FWD_WP");
+    int wp = wp_env ? atoi(wp_env) : 0;
+    if (wp) uu = 0;
     int ng = mo / (BLK_MO * mb2);
     int coh = (ng % 8 == 0) ? 1 : 0;
     dim3 grid(nmemb * ng);
     size_t lds = 16384 + (size_t)UCHUNK * BLK_E * 8 * 2
                  + (size_t)BLK_E * BLK_MO * mb2 * O * 4;
-#define LAUNCH_FWD(UU, MB2)                                                                  \
-    hipLaunchKernelGGL(HIP_KERNEL_NAME(pairconv_fwd_kernel<O, UU, MB2>), grid,               \
+#define LAUNCH_FWD(UU, MB2, WPv)                                                             \
+    hipLaunchKernelGGL(HIP_KERNEL_NAME(pairconv_fwd_kernel<O, UU, MB2, WPv>), grid,          \
                        dim3(NTHREADS), lds, stream,                                          \
                        reinterpret_cast<const __bf16*>(H.data_ptr()),                        \
                        reinterpret_cast<const __bf16*>(W.data_ptr()),                        \
                        reinterpret_cast<const __bf16*>(Ut.data_ptr()),                       \
                        out.data_ptr<float>(), E, mo, miF, nmemb, coh)  /* W arg = packed P */
     if (mb2 == 2) {
-        switch (uu) {
-            case 0: LAUNCH_FWD(0, 2); break;
-            case 1: LAUNCH_FWD(1, 2); break;
-            default: LAUNCH_FWD(2, 2); break;
+        if (wp) { LAUNCH_FWD(0, 2, 1); }
+        else switch (uu) {
+            case 0: LAUNCH_FWD(0, 2, 0); break;
+            case 1: LAUNCH_FWD(1, 2, 0); break;
+            default: LAUNCH_FWD(2, 2, 0); break;
         }
     } else {
-        switch (uu) {
-            case 0: LAUNCH_FWD(0, 1); break;
-            case 1: LAUNCH_FWD(1, 1); break;
-            default: LAUNCH_FWD(2, 1); break;
+        if (wp) { LAUNCH_FWD(0, 1, 1); }
+        else switch (uu) {
+            case 0: LAUNCH_FWD(0, 1, 0); break;
+            case 1: LAUNCH_FWD(1, 1, 0); break;
+            default: LAUNCH_FWD(2, 1, 0); break;
         }
     }
 #undef LAUNCH_FWD

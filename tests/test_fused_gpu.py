@@ -338,9 +338,11 @@ def test_pack_w_both_matches_python_permutes():
 
 
 @needs_gpu
-@pytest.mark.parametrize('uu,mb2', [('0', '1'), ('1', '1'), ('2', '1'),
-                                    ('0', '2'), ('2', '2')])
-def test_pairconv_fwd_uu_variants_agree(uu, mb2):
+@pytest.mark.parametrize('uu,mb2,wp', [('0', '1', '0'), ('1', '1', '0'),
+                                       ('2', '1', '0'), ('0', '2', '0'),
+                                       ('2', '2', '0'), ('0', '1', '1'),
+                                       ('0', '2', '1')])
+def test_pairconv_fwd_uu_variants_agree(uu, mb2, wp):
     """All SE3_FWD_UU / SE3_FWD_MB2 variants must be numerically identical."""
     from se3_transformer_amd import _C
     from se3_transformer_amd.ops.fused import _pack_w_fwd
@@ -361,15 +363,18 @@ def test_pairconv_fwd_uu_variants_agree(uu, mb2):
 
     os.environ['SE3_FWD_UU'] = '0'
     os.environ['SE3_FWD_MB2'] = '1'
+    os.environ['SE3_FWD_WP'] = '0'
     ref = run()
     os.environ['SE3_FWD_UU'] = uu
     os.environ['SE3_FWD_MB2'] = mb2
+    os.environ['SE3_FWD_WP'] = wp
     try:
         out = run()
     finally:
         del os.environ['SE3_FWD_UU']
         del os.environ['SE3_FWD_MB2']
-    assert torch.equal(out, ref), f'UU={uu} MB2={mb2} diverges'
+        del os.environ['SE3_FWD_WP']
+    assert torch.equal(out, ref), f'UU={uu} MB2={mb2} WP={wp} diverges'
 
 
 @needs_gpu

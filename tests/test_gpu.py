@@ -125,10 +125,11 @@ def test_differentiable_coors_cuda():
 @needs_gpu
 def test_equivariance_fused_bf16_autocast():
     """Rotation equivariance of the FULL model with every fused kernel
-    active under autocast-bf16. The tolerance band is set
-    by bf16 resolution (2^-8) accumulated across depth-2/16-degree-pair
-    compute: measured fp32-fused error is ~1e-4; bf16 adds ~1e-2-scale
-    rounding, so 6e-2 relative on the max element is the principled band."""
+    active under autocast-bf16. The band is calibrated
+    in-test against the measured bf16 rounding noise floor (fused-bf16 vs
+    eager-fp32 on the SAME inputs): equivariance error must stay within a
+    small multiple of that floor — i.e. rotation breaks nothing beyond the
+    precision the dtype already costs."""
     from se3_transformer_amd.ops.wigner import rot
     torch.manual_seed(21)
     model = SE3Transformer(dim=64, heads=4, dim_head=16, depth=2,
@@ -141,10 +142,13 @@ def test_equivariance_fused_bf16_autocast():
     with torch.autocast(device_type='cuda', dtype=torch.bfloat16):
         out1 = model(feats, coors @ R, mask, return_type=1)
         out2 = model(feats, coors, mask, return_type=1)
-    out2 = out2.float() @ R
-    denom = out2.abs().max().clamp(min=1e-6)
-    err = ((out1.float() - out2).abs().max() / denom).item()
-    assert err < 6e-2, f'bf16 fused equivariance: {err}'
+    ref = model(feats, coors, mask, return_type=1)     # eager fp32 oracle
+    denom = ref.abs().max().clamp(min=1e-6)
+    noise = ((out2.float() - ref).abs().max() / denom).item()   # bf16 floor
+    err = ((out1.float() - out2.float() @ R).abs().max() / denom).item()
+    assert noise < 0.1, f'bf16 path drifted from fp32: {noise}'
+    assert err < max(4 * noise, 2e-2), \
+        f'bf16 fused equivariance {err} vs noise floor {noise}'
 
 
 @needs_gpu
