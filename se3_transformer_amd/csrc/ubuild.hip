@@ -26,32 +26,45 @@ typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 #define UB_EB 16         // edges per block (bwd)
 #define UB_CB 8          // channel chunk (bwd)
 
+typedef __attribute__((ext_vector_type(2))) __bf16 ub_bf16x2;
+
 template <typename TX>
 __global__ void __launch_bounds__(UB_NT)
 ubuild_fwd_kernel(const float* __restrict__ B,   // (E, O, I, F)
                   const TX* __restrict__ X,      // (E, C, I)
                   __bf16* __restrict__ Ut,       // (C*F, O, E)
                   int E, int C, int O, int I, int F) {
+    // b and x tiles are stored [.., i PADDED to 8] bf16 so each output
+    // element is two 16B LDS reads + 4 v_dot2_f32_bf16 over i
     extern __shared__ __attribute__((aligned(16))) char smem[];
-    float* b_lds = reinterpret_cast<float*>(smem);            // [32e][O*I*F]
-    float* x_lds = b_lds + UB_E * O * I * F;                  // [32c][32e][I]
+    __bf16* b_lds = reinterpret_cast<__bf16*>(smem);          // [32e][O*F][8i]
+    __bf16* x_lds = b_lds + UB_E * O * F * 8;                 // [32c][32e][8i]
 
     const int tid = threadIdx.x;
     const int e0 = blockIdx.x * UB_E;
     const int oif = O * I * F;
+    const int of = O * F;
 
+    for (int t = tid; t < UB_E * of; t += UB_NT) {   // zero-init b pad lanes
+        *reinterpret_cast<bf16x8*>(b_lds + (size_t)t * 8) = bf16x8(0);
+    }
+    __syncthreads();
     for (int t = tid; t < UB_E * oif; t += UB_NT) {
         int e = t / oif, r = t % oif;
-        b_lds[e * oif + r] = (e0 + e < E) ? B[(size_t)(e0 + e) * oif + r] : 0.f;
+        int o = r / (I * F), rem = r % (I * F);
+        int i = rem / F, f = rem % F;
+        b_lds[((size_t)e * of + o * F + f) * 8 + i] =
+            (__bf16)((e0 + e < E) ? B[(size_t)(e0 + e) * oif + r] : 0.f);
     }
 
     for (int c0 = 0; c0 < C; c0 += UB_C) {
         __syncthreads();   // previous chunk's compute done with x_lds
-        for (int t = tid; t < UB_C * UB_E * I; t += UB_NT) {
-            int c = t / (UB_E * I), rem = t % (UB_E * I);
-            int e = rem / I, i = rem % I;
-            x_lds[(c * UB_E + e) * I + i] =
-                (e0 + e < E) ? (float)X[((size_t)(e0 + e) * C + c0 + c) * I + i] : 0.f;
+        for (int t = tid; t < UB_C * UB_E * 8; t += UB_NT) {
+            int c = t / (UB_E * 8), rem = t % (UB_E * 8);
+            int e = rem / 8, i = rem % 8;
+            x_lds[((size_t)c * UB_E + e) * 8 + i] = (__bf16)(
+                (i < I && e0 + e < E)
+                    ? (float)X[((size_t)(e0 + e) * C + c0 + c) * I + i] : 0.f);
         }
         __syncthreads();
         // out elements of this chunk: (c 32) x (f F) x (o O) x (e 32), e minor
@@ -62,11 +75,16 @@ ubuild_fwd_kernel(const float* __restrict__ B,   // (E, O, I, F)
             int o = r % O;
             int cf = r / O;
             int f = cf % F, c = cf / F;
-            const float* brow = b_lds + e * oif + (o * I) * F + f;
-            const float* xrow = x_lds + (c * UB_E + e) * I;
+            bf16x8 bv = *reinterpret_cast<const bf16x8*>(
+                b_lds + ((size_t)e * of + o * F + f) * 8);
+            bf16x8 xv = *reinterpret_cast<const bf16x8*>(
+                x_lds + ((size_t)c * UB_E + e) * 8);
+            const ub_bf16x2* b2 = reinterpret_cast<const ub_bf16x2*>(&bv);
+            const ub_bf16x2* x2 = reinterpret_cast<const ub_bf16x2*>(&xv);
             float acc = 0.f;
-            for (int i = 0; i < I; ++i)
-                acc = fmaf(brow[(size_t)i * F], xrow[i], acc);
+#pragma unroll
+            for (int p = 0; p < 4; ++p)
+                acc = __builtin_amdgcn_fdot2_f32_bf16(b2[p], x2[p], acc, false);
             if (e0 + e < E)
                 Ut[((size_t)((c0 + c) * F + f) * O + o) * E + e0 + e] = (__bf16)acc;
         }
@@ -131,9 +149,10 @@ void ubuild_fwd(torch::Tensor B, torch::Tensor X, torch::Tensor Ut,
     int C = X.size(1);
     TORCH_CHECK(C % UB_C == 0, "channels must be a multiple of 32");
     TORCH_CHECK(O * I * F <= 343, "degree pair too large for LDS staging");
+    TORCH_CHECK(I <= 8, "I (2*d_in+1) must be <= 8");
     auto stream = at::cuda::getCurrentHIPStream();
     dim3 grid((E + UB_E - 1) / UB_E);
-    size_t lds = (size_t)UB_E * O * I * F * 4 + (size_t)UB_C * UB_E * I * 4;
+    size_t lds = (size_t)UB_E * O * F * 8 * 2 + (size_t)UB_C * UB_E * 8 * 2;
     if (X.dtype() == torch::kFloat32) {
         hipLaunchKernelGGL(HIP_KERNEL_NAME(ubuild_fwd_kernel<float>), grid,
                            dim3(UB_NT), lds, stream, B.data_ptr<float>(),
