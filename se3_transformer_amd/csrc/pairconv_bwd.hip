@@ -370,7 +370,8 @@ pairconv_bwd_dw_kernel(const __bf16* __restrict__ Gt,  // (mo, O, E)
 // block: 64 e x 32 urow, loops mo in blocks of 8; R tile recomputed by MFMA
 // exactly as the forward kernel, bounced through LDS, contracted vs g.
 // ---------------------------------------------------------------------------
-template <int O>
+template <int O, int WP>   // WP: prefetch next mo-block's W frags under the
+                           // contraction phase (see pairconv.hip fwd)
 __global__ void __launch_bounds__(NT)
 pairconv_bwd_du_kernel(const __bf16* __restrict__ H,   // (E,128)
                        const __bf16* __restrict__ P,   // packed W as forward
@@ -441,9 +442,20 @@ pairconv_bwd_du_kernel(const __bf16* __restrict__ H,   // (E,128)
         }
     };
     load_gb(0);
-    __syncthreads();
 
     const int nmo = mo / 8;
+    auto pb_of = [&](int mbb) {
+        return P + ((((size_t)mbb * (miF / 32) + cb) * 4 + wm) * 4) * 4 * 64 * 8
+               + (size_t)lane * 8;
+    };
+    bf16x8 a_pre[WP ? 16 : 1];
+    if (WP) {
+#pragma unroll
+        for (int f = 0; f < 16; ++f)
+            a_pre[f] = *reinterpret_cast<const bf16x8*>(pb_of(0) + (size_t)f * 512);
+    }
+    __syncthreads();
+
     for (int mb = 0; mb < nmo; ++mb) {
         // commit the staged g tile [8][O][64] + bias chunk [256]
         if (tid < GTOT) {
@@ -459,15 +471,16 @@ pairconv_bwd_du_kernel(const __bf16* __restrict__ H,   // (E,128)
         for (int mf = 0; mf < 4; ++mf)
 #pragma unroll
             for (int ef = 0; ef < 2; ++ef) acc[mf][ef] = {0.f, 0.f, 0.f, 0.f};
-        const __bf16* pbase = P + ((((size_t)mb * (miF / 32) + cb) * 4 + wm) * 4) * 4 * 64 * 8
-                              + (size_t)lane * 8;
+        const __bf16* pbase = pb_of(mb);
 #pragma unroll
         for (int kit = 0; kit < 4; ++kit) {
             const int k0 = kit * 32 + l4 * 8;
             bf16x8 a[4], b[2];
 #pragma unroll
             for (int mf = 0; mf < 4; ++mf) {
-                a[mf] = *reinterpret_cast<const bf16x8*>(pbase + ((size_t)mf * 4 + kit) * 64 * 8);
+                a[mf] = WP ? a_pre[mf * 4 + kit]
+                           : *reinterpret_cast<const bf16x8*>(
+                                 pbase + ((size_t)mf * 4 + kit) * 64 * 8);
             }
 #pragma unroll
             for (int ef = 0; ef < 2; ++ef) {
@@ -480,6 +493,13 @@ pairconv_bwd_du_kernel(const __bf16* __restrict__ H,   // (E,128)
 #pragma unroll
                 for (int ef = 0; ef < 2; ++ef)
                     acc[mf][ef] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[mf], b[ef], acc[mf][ef], 0, 0, 0);
+        }
+        if (WP) {   // issue next mo-block's fragment loads under the
+                    // R-bounce + contraction phases below
+            const __bf16* pn = pb_of(mb + 1 < nmo ? mb + 1 : mb);
+#pragma unroll
+            for (int f = 0; f < 16; ++f)
+                a_pre[f] = *reinterpret_cast<const bf16x8*>(pn + (size_t)f * 512);
         }
         // bounce R (+bias) to LDS [256n][64e]
 #pragma unroll
