@@ -372,7 +372,7 @@ pairconv_bwd_dw_kernel(const __bf16* __restrict__ Gt,  // (mo, O, E)
 // ---------------------------------------------------------------------------
 template <int O, int WP>   // WP: prefetch next mo-block's W frags under the
                            // contraction phase (see pairconv.hip fwd)
-__global__ void __launch_bounds__(NT)
+__global__ void __launch_bounds__(NT, 4)   // keep 2 blocks/CU under WP
 pairconv_bwd_du_kernel(const __bf16* __restrict__ H,   // (E,128)
                        const __bf16* __restrict__ P,   // packed W as forward
                        const float* __restrict__ bias, // (mo*miF,)
@@ -608,15 +608,31 @@ void pairconv_bwd_du(torch::Tensor H, torch::Tensor W, torch::Tensor bias,
     int ncb = miF / 32;
     int coh = (ncb % 8 == 0) ? 1 : 0;
     dim3 grid(nmemb * ncb);
+    // WP off by default here: du's contraction phase holds more live state
+    // than fwd's epilogue, so the prefetch spills 116-240 B/lane under the
+    // 2-blocks/CU cap (vs fwd's 24-36 B at O=3/5). SE3_DU_WP=1 for A/B.
+    const char* wp_env = getenv("SE3_DU_WP");
     DISPATCH_O(O, {
+        int wp = wp_env ? atoi(wp_env) : 0;
         size_t lds = 49152 + (size_t)((8 * kO * 64 * 2 + 15) & ~15) +
                      (size_t)32 * kO * 64 * 4 + 256 * 4;
-        hipLaunchKernelGGL(HIP_KERNEL_NAME(pairconv_bwd_du_kernel<kO>), grid, dim3(NT), lds, stream,
-                           reinterpret_cast<const __bf16*>(H.data_ptr()),
-                           reinterpret_cast<const __bf16*>(W.data_ptr()),
-                           bias.data_ptr<float>(),
-                           reinterpret_cast<const __bf16*>(Gt.data_ptr()),
-                           dU.data_ptr<float>(), E, mo, miF, nmemb, coh);
+        if (wp) {
+            hipLaunchKernelGGL(HIP_KERNEL_NAME(pairconv_bwd_du_kernel<kO, 1>),
+                               grid, dim3(NT), lds, stream,
+                               reinterpret_cast<const __bf16*>(H.data_ptr()),
+                               reinterpret_cast<const __bf16*>(W.data_ptr()),
+                               bias.data_ptr<float>(),
+                               reinterpret_cast<const __bf16*>(Gt.data_ptr()),
+                               dU.data_ptr<float>(), E, mo, miF, nmemb, coh);
+        } else {
+            hipLaunchKernelGGL(HIP_KERNEL_NAME(pairconv_bwd_du_kernel<kO, 0>),
+                               grid, dim3(NT), lds, stream,
+                               reinterpret_cast<const __bf16*>(H.data_ptr()),
+                               reinterpret_cast<const __bf16*>(W.data_ptr()),
+                               bias.data_ptr<float>(),
+                               reinterpret_cast<const __bf16*>(Gt.data_ptr()),
+                               dU.data_ptr<float>(), E, mo, miF, nmemb, coh);
+        }
     });
     hipError_t err = hipGetLastError();
     TORCH_CHECK(err == hipSuccess, "bwd_du: ", hipGetErrorString(err));

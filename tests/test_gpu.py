@@ -139,8 +139,12 @@ def test_equivariance_fused_bf16_autocast():
     coors = torch.randn(1, 48, 3, device='cuda') * 1.5
     mask = torch.ones(1, 48, dtype=torch.bool, device='cuda')
     R = rot(23., 117., 195.).float().to('cuda')
+    # rotate OUTSIDE autocast: under autocast the matmul would round the
+    # rotated coordinates to bf16, perturbing the k-NN selection itself
+    # (discrete neighbor flips, not a kernel equivariance failure)
+    coors_rot = coors @ R
     with torch.autocast(device_type='cuda', dtype=torch.bfloat16):
-        out1 = model(feats, coors @ R, mask, return_type=1)
+        out1 = model(feats, coors_rot, mask, return_type=1)
         out2 = model(feats, coors, mask, return_type=1)
     ref = model(feats, coors, mask, return_type=1)     # eager fp32 oracle
     denom = ref.abs().max().clamp(min=1e-6)
