@@ -16,7 +16,21 @@ from torch import nn
 from ..utils import batched_index_select, map_values, to_order
 from .core import ConvSE3, LinearSE3, NormSE3, ResidualSE3
 from .fiber import Fiber
-from .rotary import apply_rotary_pos_emb
+
+
+def apply_rotary_pos_emb(t, freqs):
+    """Eager rotary rotation over the leading `rot_dim` feature rows — the
+    fallback/oracle for the in-kernel rotation (csrc/attn2.hip rope_apply;
+    semantics of reference rotary.py:15-24). `t` is (..., d, m); `freqs`
+    broadcasts over everything but the d axis."""
+    rot_dim = freqs.shape[-2]
+    t_rot, t_pass = t[..., :rot_dim, :], t[..., rot_dim:, :]
+    # pair layout (2i, 2i+1) -> halves (-odd, even), as the reference's
+    # rotate_half does (and as rope_P reproduces lane-wise in the kernel)
+    even, odd = t_rot[..., 0::2, :], t_rot[..., 1::2, :]
+    half_rot = torch.cat((-odd, even), dim=-2)
+    t_rot = t_rot * freqs.cos() + half_rot * freqs.sin()
+    return torch.cat((t_rot, t_pass), dim=-2)
 
 
 def _first_tensor(features):

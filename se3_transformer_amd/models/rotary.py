@@ -1,6 +1,11 @@
-"""Rotary positional embeddings applied to degree-0 attention channels.
+"""Rotary positional embedding tables for degree-0 attention channels.
 
-Semantics parity with reference rotary.py:1-24.
+On the MI355X path the rotation itself happens INSIDE the attention kernel
+(csrc/attn2.hip rotates q/k/v in-registers from the raw frequency tables),
+so this module is only the host-side table builder. The eager fallback
+rotation lives next to its consumer in models/attention.py
+(`apply_rotary_pos_emb`, re-exported here for reference API parity with
+rotary.py:1-24).
 """
 from __future__ import annotations
 
@@ -9,6 +14,9 @@ from torch import nn
 
 
 class SinusoidalEmbeddings(nn.Module):
+    """Frequency table: t -> (…, dim) with pairwise-repeated inverse
+    frequencies (reference rotary.py:5-13)."""
+
     def __init__(self, dim):
         super().__init__()
         inv_freq = 1. / (10000 ** (torch.arange(0, dim, 2).float() / dim))
@@ -19,16 +27,6 @@ class SinusoidalEmbeddings(nn.Module):
         return freqs.repeat_interleave(2, dim=-1)
 
 
-def rotate_half(x):
-    # x: [..., d, m] with d even: pairs (x1, x2) -> (-x2, x1)
-    shape = x.shape
-    x = x.view(*shape[:-2], shape[-2] // 2, 2, shape[-1])
-    x1, x2 = x.unbind(dim=-2)
-    return torch.cat((-x2, x1), dim=-2)
-
-
 def apply_rotary_pos_emb(t, freqs):
-    rot_dim = freqs.shape[-2]
-    t, t_pass = t[..., :rot_dim, :], t[..., rot_dim:, :]
-    t = (t * freqs.cos()) + (rotate_half(t) * freqs.sin())
-    return torch.cat((t, t_pass), dim=-2)
+    from .attention import apply_rotary_pos_emb as _impl
+    return _impl(t, freqs)
