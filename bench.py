@@ -84,6 +84,10 @@ def main():
     p.add_argument('--device', type=str, default=None)
     p.add_argument('--timers', action='store_true',
                    help='report per-phase (fwd/bwd/opt) hipEvent timings')
+    p.add_argument('--grad-compression', default='none',
+                   choices=['none', 'bf16'],
+                   help='DP gradient all-reduce dtype (bf16 halves xGMI '
+                        'traffic; see parallel/ddp.py for the error bound)')
     p.add_argument('--preset', default=None, choices=sorted(PRESETS),
                    help='one of the BASELINE.json named configs')
     args = p.parse_args()
@@ -106,7 +110,9 @@ def main():
     model = build_model(args, device)
     n_params = sum(p_.numel() for p_ in model.parameters())
 
-    ddp = DistributedDataParallelSE3(model, sync_params=False) if world > 1 else None
+    ddp = DistributedDataParallelSE3(
+        model, sync_params=False,
+        grad_compression=args.grad_compression) if world > 1 else None
     runner = ddp if ddp is not None else model
 
     opt = torch.optim.SGD(model.parameters(), lr=1e-4)
@@ -165,10 +171,13 @@ def main():
               file=sys.stderr, flush=True)
 
     # hipGraph capture: the step is shape-static, so capture once and replay —
-    # removes the Python/launch-gap overhead between the ~10k kernels per step.
-    # Default on for single-GPU; for world > 1 pass --graph explicitly (RCCL
-    # collectives are stream-captured, but multi-rank capture is not validated
-    # on this RCCL build, and a capture hang would kill a scaling run).
+    # removes the Python/launch-gap overhead between the thousands of kernels
+    # per step. Default on for single-GPU. For world > 1 pass --graph
+    # explicitly: capture of an RCCL collective is VALIDATED at 1 rank
+    # (tests/test_gpu_nccl.py::test_hipgraph_capture_of_rccl_allreduce), but
+    # multi-rank capture has not run on real multi-GPU hardware yet and a
+    # capture hang would kill a scaling run — the eager fallback costs ~15%
+    # and is first-try safe.
     graph_mode = (args.graph if args.graph is not None else world == 1) \
         and use_cuda
     if graph_mode:
