@@ -469,7 +469,8 @@ def test_attn2_module_vs_eager_with_backward(kn):
 @needs_gpu
 def test_attn2_rotary_in_kernel_vs_eager():
     """Rotary q/k/v rotation folded into attn2 vs the eager apply_rotary
-    path, full model forward (fp32)."""
+    path, full model forward AND backward (the in-kernel rotation's
+    transpose map feeds dq/dk/dv), fp32."""
     from se3_transformer_amd import SE3Transformer
     torch.manual_seed(10)
     model = SE3Transformer(dim=32, heads=2, dim_head=16, depth=1,
@@ -481,11 +482,21 @@ def test_attn2_rotary_in_kernel_vs_eager():
     os.environ['SE3_EAGER_ATTN'] = '1'
     try:
         ref = model(feats, coors, mask, return_type=0)
+        ref.pow(2).mean().backward()
     finally:
         del os.environ['SE3_EAGER_ATTN']
+    gref = {n: p.grad.clone() for n, p in model.named_parameters()
+            if p.grad is not None}
+    model.zero_grad()
     out = model(feats, coors, mask, return_type=0)
     err = _rel_err(out, ref)
     assert err < 2e-3, f'rotary-in-kernel parity: {err}'
+    out.pow(2).mean().backward()
+    for n, p in model.named_parameters():
+        if n not in gref:
+            continue
+        gerr = _rel_err(p.grad.float(), gref[n].float())
+        assert gerr < 5e-3, f'rotary-in-kernel grad {n}: {gerr}'
 
 
 @needs_gpu
