@@ -448,6 +448,16 @@ void ubuild_fwd(torch::Tensor B, torch::Tensor X, torch::Tensor Ut,
                 int64_t O, int64_t I, int64_t F);
 void ubuild_bwd_dx(torch::Tensor B, torch::Tensor dU, torch::Tensor dX,
                    int64_t O, int64_t I, int64_t F);
+void egnn_rel_dist_fwd(torch::Tensor ht, torch::Tensor idx, torch::Tensor dist);
+void egnn_rel_dist_bwd(torch::Tensor ht, torch::Tensor idx, torch::Tensor gdist,
+                       torch::Tensor dht);
+void egnn_htype_update_fwd(torch::Tensor ht, torch::Tensor idx, torch::Tensor w,
+                           torch::Tensor sc, torch::Tensor bi,
+                           torch::Tensor upd, double eps);
+void egnn_htype_update_bwd(torch::Tensor ht, torch::Tensor idx, torch::Tensor w,
+                           torch::Tensor sc, torch::Tensor bi, torch::Tensor g,
+                           torch::Tensor dht, torch::Tensor dw,
+                           torch::Tensor dsc, torch::Tensor dbi, double eps);
 void radial_trunk_fwd(torch::Tensor X, torch::Tensor W0, torch::Tensor p0,
                       torch::Tensor W3, torch::Tensor p3,
                       torch::Tensor H, torch::Tensor yh0, torch::Tensor yh3,
@@ -482,4 +492,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("ubuild_fwd", &ubuild_fwd,
           "basis x features precontraction (e-contiguous out)");
     m.def("ubuild_bwd_dx", &ubuild_bwd_dx, "ubuild dX backward");
+    m.def("egnn_rel_dist_fwd", &egnn_rel_dist_fwd,
+          "EGNN neighbor rel-htype distances (no n^2 intermediate)");
+    m.def("egnn_rel_dist_bwd", &egnn_rel_dist_bwd, "its backward");
+    m.def("egnn_htype_update_fwd", &egnn_htype_update_fwd,
+          "EGNN htype norm+weighted neighbor sum (no n^2 intermediate)");
+    m.def("egnn_htype_update_bwd", &egnn_htype_update_bwd, "its backward");
 }

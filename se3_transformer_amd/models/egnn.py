@@ -75,19 +75,35 @@ class EGNN(nn.Module):
         htype_degrees = [deg for deg, _ in htypes]
         htype_dims = [t.shape[-2] for _, t in htypes]
 
-        # relative higher-type differences and their norms
+        # MI355X path (csrc/egnn.hip): rel-htype distances and the
+        # norm+weighted-sum update are computed straight from the gathered
+        # neighbor indices — the eager path's O(n^2 d m) pairwise rel
+        # tensor (reference :801-836) never exists, which is what makes
+        # the EGNN trunk usable at n ~ 1024.
+        from ..ops import fused as _fused
+        max_m = max((t.shape[-1] for _, t in htypes), default=0)
+        use_kernels = (nodes.is_cuda and len(htypes) > 0
+                       and _fused.egnn_kernels_ok(max_m))
+
         rel_htypes = []
-        rel_htypes_dists = []
-        for _, htype in htypes:
-            rel_htype = htype.unsqueeze(2) - htype.unsqueeze(1)  # b i j d m
-            rel_htypes.append(rel_htype)
-            rel_htypes_dists.append(rel_htype.norm(dim=-1))
+        if use_kernels:
+            neighbor_htype_dists = [
+                _fused.htype_rel_dist(t, neighbor_indices).to(t.dtype)
+                for _, t in htypes]
+        else:
+            # relative higher-type differences and their norms (eager)
+            rel_htypes_dists = []
+            for _, htype in htypes:
+                rel_htype = htype.unsqueeze(2) - htype.unsqueeze(1)  # b i j d m
+                rel_htypes.append(rel_htype)
+                rel_htypes_dists.append(rel_htype.norm(dim=-1))
+            neighbor_htype_dists = [
+                batched_index_select(t, neighbor_indices, dim=2)
+                for t in rel_htypes_dists]
 
         # edge MLP inputs
         nodes_i = nodes.unsqueeze(2)  # b i 1 d
         nodes_j = batched_index_select(nodes, neighbor_indices, dim=1)  # b i k d
-        neighbor_htype_dists = [batched_index_select(t, neighbor_indices, dim=2)
-                                for t in rel_htypes_dists]
         coor_rel_dist = rel_dist.unsqueeze(-1)  # b i j 1
 
         edge_mlp_inputs = broadcat((nodes_i, nodes_j, *neighbor_htype_dists, coor_rel_dist), dim=-1)
@@ -108,10 +124,19 @@ class EGNN(nn.Module):
         split_htype_weights = htype_weights.split(htype_dims, dim=-1)
 
         htype_updates = []
-        for degree, rel_htype, htype_weight in zip(htype_degrees, rel_htypes, split_htype_weights):
-            normed_rel_htype = self.htype_norms[str(degree)](rel_htype)
-            normed_rel_htype = batched_index_select(normed_rel_htype, neighbor_indices, dim=2)
-            htype_updates.append(torch.einsum('bijdm,bijd->bidm', normed_rel_htype, htype_weight))
+        if use_kernels:
+            for degree, (_, htype), htype_weight in zip(
+                    htype_degrees, htypes, split_htype_weights):
+                norm_mod = self.htype_norms[str(degree)]
+                upd = _fused.htype_update(htype, neighbor_indices,
+                                          htype_weight, norm_mod.scale,
+                                          norm_mod.bias, norm_mod.eps)
+                htype_updates.append(upd.to(htype.dtype))
+        else:
+            for degree, rel_htype, htype_weight in zip(htype_degrees, rel_htypes, split_htype_weights):
+                normed_rel_htype = self.htype_norms[str(degree)](rel_htype)
+                normed_rel_htype = batched_index_select(normed_rel_htype, neighbor_indices, dim=2)
+                htype_updates.append(torch.einsum('bijdm,bijd->bidm', normed_rel_htype, htype_weight))
 
         # node updates
         if mask is not None:

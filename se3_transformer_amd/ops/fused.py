@@ -323,6 +323,81 @@ def ubuild(x, B, O, I, F):
     return _UBuildFn.apply(x, B, O, I, F)
 
 
+class _HtypeRelDistFn(torch.autograd.Function):
+    """EGNN neighbor rel-htype distances (csrc/egnn.hip): straight from the
+    gathered neighbor indices — the O(n^2 d m) pairwise rel tensor of the
+    eager path never exists."""
+
+    @staticmethod
+    def forward(ctx, ht, idx):
+        ext = _load_ext()
+        b, n, d, m = ht.shape
+        k = idx.shape[2]
+        dist = torch.empty(b, n, k, d, dtype=torch.float32, device=ht.device)
+        htc = ht.contiguous()
+        ext.egnn_rel_dist_fwd(htc, idx.contiguous(), dist)
+        ctx.save_for_backward(htc, idx)
+        return dist
+
+    @staticmethod
+    def backward(ctx, gdist):
+        ext = _load_ext()
+        ht, idx = ctx.saved_tensors
+        dht = torch.zeros(ht.shape, dtype=torch.float32, device=ht.device)
+        ext.egnn_rel_dist_bwd(ht, idx, gdist.contiguous().float(), dht)
+        return dht.to(ht.dtype), None
+
+
+def htype_rel_dist(ht, idx):
+    return _HtypeRelDistFn.apply(ht, idx)
+
+
+class _HtypeUpdateFn(torch.autograd.Function):
+    """EGNN htype update (csrc/egnn.hip): HtypesNorm of the neighbor
+    rel-htypes + weighted sum over neighbors, fused and gather-based."""
+
+    @staticmethod
+    def forward(ctx, ht, idx, w, scale, bias, eps):
+        ext = _load_ext()
+        b, n, d, m = ht.shape
+        htc = ht.contiguous()
+        wc = w.detach().float().contiguous()
+        s32 = scale.detach().reshape(-1).float().contiguous()
+        b32 = bias.detach().reshape(-1).float().contiguous()
+        upd = torch.empty(b, n, d, m, dtype=torch.float32, device=ht.device)
+        ext.egnn_htype_update_fwd(htc, idx.contiguous(), wc, s32, b32,
+                                  upd, eps)
+        ctx.save_for_backward(htc, idx, wc, s32, b32)
+        ctx.meta = (eps, w.dtype, scale.shape, scale.dtype)
+        return upd
+
+    @staticmethod
+    def backward(ctx, g):
+        ext = _load_ext()
+        ht, idx, wc, s32, b32 = ctx.saved_tensors
+        eps, w_dtype, p_shape, p_dtype = ctx.meta
+        dht = torch.zeros(ht.shape, dtype=torch.float32, device=ht.device)
+        dw = torch.empty(wc.shape, dtype=torch.float32, device=ht.device)
+        dsc = torch.zeros_like(s32)
+        dbi = torch.zeros_like(b32)
+        ext.egnn_htype_update_bwd(ht, idx, wc, s32, b32,
+                                  g.contiguous().float(), dht, dw, dsc, dbi,
+                                  eps)
+        return (dht.to(ht.dtype), None, dw.to(w_dtype),
+                dsc.view(p_shape).to(p_dtype), dbi.view(p_shape).to(p_dtype),
+                None)
+
+
+def htype_update(ht, idx, w, scale, bias, eps):
+    return _HtypeUpdateFn.apply(ht, idx, w, scale, bias, eps)
+
+
+def egnn_kernels_ok(m: int) -> bool:
+    ext = _load_ext()
+    return (ext is not None and hasattr(ext, 'egnn_htype_update_fwd')
+            and m <= 7 and os.environ.get('SE3_EAGER_EGNN') != '1')
+
+
 RADIAL_TRUNK_DIMS = (1, 2, 3, 9, 17)   # instantiated in csrc/radial.hip
 
 

@@ -28,6 +28,7 @@ setup(
                      'se3_transformer_amd/csrc/pack_w.hip',
                      'se3_transformer_amd/csrc/radial.hip',
                      'se3_transformer_amd/csrc/ubuild.hip',
+                     'se3_transformer_amd/csrc/egnn.hip',
                      'se3_transformer_amd/csrc/sh_basis.hip',
                      'se3_transformer_amd/csrc/norm_se3.hip',
                      'se3_transformer_amd/csrc/attn2.hip',
