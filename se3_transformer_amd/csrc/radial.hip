@@ -407,13 +407,17 @@ radial_trunk_bwd_kernel(const __bf16* __restrict__ dHg,  // (E, 128)
         case 1: { constexpr int kD = 1; __VA_ARGS__; break; }       \
         case 2: { constexpr int kD = 2; __VA_ARGS__; break; }       \
         case 3: { constexpr int kD = 3; __VA_ARGS__; break; }       \
+        case 5: { constexpr int kD = 5; __VA_ARGS__; break; }       \
         case 9: { constexpr int kD = 9; __VA_ARGS__; break; }       \
         case 17: { constexpr int kD = 17; __VA_ARGS__; break; }     \
+        case 21: { constexpr int kD = 21; __VA_ARGS__; break; }     \
+        case 25: { constexpr int kD = 25; __VA_ARGS__; break; }     \
         default: TORCH_CHECK(false, "unsupported edge dim ", D);    \
     }
 
 bool radial_trunk_dim_ok(int64_t d) {
-    return d == 1 || d == 2 || d == 3 || d == 9 || d == 17;
+    return d == 1 || d == 2 || d == 3 || d == 5 || d == 9 || d == 17 ||
+           d == 21 || d == 25;
 }
 
 void radial_trunk_fwd(torch::Tensor X, torch::Tensor W0, torch::Tensor p0,

@@ -398,7 +398,7 @@ def egnn_kernels_ok(m: int) -> bool:
             and m <= 7 and os.environ.get('SE3_EAGER_EGNN') != '1')
 
 
-RADIAL_TRUNK_DIMS = (1, 2, 3, 9, 17)   # instantiated in csrc/radial.hip
+RADIAL_TRUNK_DIMS = (1, 2, 3, 5, 9, 17, 21, 25)   # instantiated in csrc/radial.hip
 
 
 def radial_trunk_ok(in_dim: int, mid_dim: int) -> bool:
