@@ -415,11 +415,6 @@ radial_trunk_bwd_kernel(const __bf16* __restrict__ dHg,  // (E, 128)
         default: TORCH_CHECK(false, "unsupported edge dim ", D);    \
     }
 
-bool radial_trunk_dim_ok(int64_t d) {
-    return d == 1 || d == 2 || d == 3 || d == 5 || d == 9 || d == 17 ||
-           d == 21 || d == 25;
-}
-
 void radial_trunk_fwd(torch::Tensor X, torch::Tensor W0, torch::Tensor p0,
                       torch::Tensor W3, torch::Tensor p3,
                       torch::Tensor H, torch::Tensor yh0, torch::Tensor yh3,
