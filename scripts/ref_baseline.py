@@ -43,9 +43,10 @@ import torch  # noqa: E402
 def stage_reference(tmp='/tmp/ref_baseline'):
     """Copy the reference package + shims into an importable temp tree.
 
-    On GPU boxes /root/reference is not mounted; the caller ships it via
-    the git-ignored .ref_stage/ snapshot directory instead (kept out of
-    the repo history — measurement input, not framework code)."""
+    On GPU boxes /root/reference is not mounted; to run there, copy it
+    into the git-ignored .ref_stage/ first (kept out of the repo history
+    and DELETED after measuring — reference code never lives in the
+    repo):  cp -r /root/reference/se3_transformer_pytorch .ref_stage/"""
     ref_src = '/root/reference/se3_transformer_pytorch'
     if not os.path.isdir(ref_src):
         ref_src = os.path.join(REPO, '.ref_stage', 'se3_transformer_pytorch')
