@@ -93,8 +93,11 @@ def main():
     args = p.parse_args()
     args.model_kwargs = {}
     if args.preset:
+        # preset values fill in wherever the user did not pass the flag
+        # explicitly (an explicit --batch etc. wins over the preset)
         for k, v in PRESETS[args.preset].items():
-            setattr(args, k, v)
+            if k == 'model_kwargs' or getattr(args, k, None) == p.get_default(k):
+                setattr(args, k, v)
 
     rank, world, local_rank = setup_distributed()
     use_cuda = torch.cuda.is_available()
