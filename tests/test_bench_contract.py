@@ -54,3 +54,21 @@ def test_denoise_example_runs():
                        capture_output=True, text=True, cwd=REPO, timeout=600)
     assert r.returncode == 0, r.stderr[-1500:]
     assert 'loss:' in r.stdout
+
+
+def test_infer_example_cpu_smoke():
+    """examples/infer.py runs end-to-end on CPU (tiny config)."""
+    import json
+    import subprocess
+    import sys
+    out = subprocess.run(
+        [sys.executable, 'examples/infer.py', '--points', '24', '--dim', '16',
+         '--heads', '2', '--dim-head', '8', '--depth', '1',
+         '--num-degrees', '2', '--num-neighbors', '4', '--iters', '1',
+         '--warmup', '0', '--no-graph'],
+        capture_output=True, text=True, timeout=300,
+        cwd=__import__('os').path.dirname(__import__('os').path.dirname(
+            __import__('os').path.abspath(__file__))))
+    assert out.returncode == 0, out.stderr[-2000:]
+    parsed = json.loads(out.stdout.strip().splitlines()[-1])
+    assert parsed['samples_per_sec'] > 0
