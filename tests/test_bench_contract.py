@@ -35,11 +35,14 @@ def test_bench_single_process():
 
 
 def test_bench_two_process_gloo():
+    # hide any GPU so setup_distributed picks gloo: two ranks cannot share
+    # one device over nccl (this test exercises the CPU DP path everywhere)
+    env = dict(os.environ, HIP_VISIBLE_DEVICES='', CUDA_VISIBLE_DEVICES='')
     r = subprocess.run(
         [sys.executable, '-m', 'torch.distributed.run', '--nnodes=1',
          '--nproc-per-node', '2', '--master-addr', '127.0.0.1',
          '--master-port', '29719', 'bench.py', '--gpus', '2'] + TINY,
-        capture_output=True, text=True, cwd=REPO, timeout=900)
+        capture_output=True, text=True, cwd=REPO, timeout=900, env=env)
     assert r.returncode == 0, r.stderr[-2000:]
     d = _check_line(r.stdout)
     assert d['n_gpus'] == 2 and d['config']['parallelism'] == 'dp2'
