@@ -41,6 +41,16 @@ def _remove_self(t, off_diag):
     return t.gather(2, idx)
 
 
+def _pairwise_edges(edges, n):
+    """Accept per-NODE edge features (b, n, d) alongside the usual pairwise
+    (b, n, n, d): the reference's masked_select at :1235 broadcasts the 3-D
+    case so the pair (i, j) edge depends on the target node j only."""
+    if edges.dim() == 3:
+        edges = edges.unsqueeze(1).expand(edges.shape[0], n, n,
+                                          edges.shape[-1])
+    return edges
+
+
 class SE3Transformer(nn.Module):
     def __init__(
         self,
@@ -338,6 +348,7 @@ class SE3Transformer(nn.Module):
             if exists(edges) and exists(self.edge_emb):
                 edges = self.edge_emb(edges)
             if exists(edges):
+                edges = _pairwise_edges(edges, n)
                 edges = batched_index_select(edges, neighbor_indices, dim=2)
             if exists(self.adj_emb):
                 # adj_indices is off-diagonal (b,n,n-1): map the selected
@@ -361,7 +372,7 @@ class SE3Transformer(nn.Module):
             if exists(edges):
                 if exists(self.edge_emb):
                     edges = self.edge_emb(edges)
-                edges = _remove_self(edges, off_diag)
+                edges = _remove_self(_pairwise_edges(edges, n), off_diag)
 
             if exists(self.adj_emb):
                 adj_emb = self.adj_emb(adj_indices)

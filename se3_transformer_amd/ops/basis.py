@@ -51,8 +51,11 @@ def num_basis_freq(d_in: int, d_out: int) -> int:
 
 
 def get_R_tensor(order_out: int, order_in: int, a, b, c) -> torch.Tensor:
-    """Kronecker product D_out(R) ⊗ D_in(R) (reference basis.py:110)."""
-    return torch.kron(wigner_d(order_out, a, b, c), wigner_d(order_in, a, b, c))
+    """Kronecker product D_out(R) ⊗ D_in(R) (reference basis.py:110).
+    Returned in the caller's default dtype, like the reference (the Q_J
+    solver runs under a float64 default-dtype context)."""
+    return torch.kron(wigner_d(order_out, a, b, c),
+                      wigner_d(order_in, a, b, c)).to(torch.get_default_dtype())
 
 
 def _null_space_1d(mats, eps=1e-9) -> torch.Tensor:

@@ -98,8 +98,10 @@ def test_wigner_is_representation():
         assert (d1 @ d1.t() - eye).abs().max() < 1e-10
 
 
+@torch_default_dtype(torch.float64)
 def test_qj_intertwiner_property():
-    """(D_out ⊗ D_in) Q_J = Q_J D_J at random angles — reference tests/test_basis.py:11."""
+    """(D_out ⊗ D_in) Q_J = Q_J D_J at random angles — reference tests/test_basis.py:11.
+    (get_R_tensor follows the default dtype, as the reference's does.)"""
     g = torch.Generator().manual_seed(11)
     rand_angles = torch.rand(4, 3, generator=g, dtype=torch.float64)
     for (J, order_in, order_out) in [(1, 1, 1), (2, 1, 1), (3, 2, 1), (2, 2, 2)]:
