@@ -61,7 +61,9 @@ __device__ __forceinline__ float bf16_to_f32(unsigned short u) {
 // (where the MFMA accumulators are dead), so each MFMA phase starts with
 // its operands already in flight instead of paying a fresh L2/HBM round
 // trip (T14 applied to the W stream).
-template <int O, int UU, int MB2, int WP>   // UU = 16B u-units register-staged per thread
+// ES = epilogue split: process the padded o range in two passes with half
+// the persistent accumulators (frees VGPRs for the WP prefetch at O>=5)
+template <int O, int UU, int MB2, int WP, int ES>   // UU = 16B u-units register-staged per thread
 __global__ void __launch_bounds__(NTHREADS, 4)   // cap VGPR<=128: 2 blocks/CU
 pairconv_fwd_kernel(const __bf16* __restrict__ H,
                     const __bf16* __restrict__ P,   // packed W: [mo/8][miF/32][wm4][mf4][kit4][lane64][8]
@@ -246,15 +248,24 @@ pairconv_fwd_kernel(const __bf16* __restrict__ H,
 
         // ---- epilogue: contract acc against u_lds into s[ef][moi][o-pairs].
         // One ds_read_b128 per (row, e) covers all 8 padded o; the o-pair
-        // float2 accumulation maps onto v_pk_fma_f32.
+        // float2 accumulation maps onto v_pk_fma_f32. With ES (epilogue
+        // split), the o range is processed in NH sequential halves with
+        // only PH o-pairs of accumulators live (re-reads u but halves the
+        // persistent s footprint — funds WP at the big orders); a
+        // sched_barrier keeps the halves from being re-merged.
         typedef __attribute__((ext_vector_type(2))) float f32x2e;
-        f32x2e s[2][2][4];
+        constexpr int PH = ES ? 2 : 4;   // o-pairs held per pass
+        constexpr int NH = 4 / PH;
+#pragma unroll
+        for (int oh = 0; oh < NH; ++oh) {
+        if (NH > 1 && oh > 0) __builtin_amdgcn_sched_barrier(0);
+        f32x2e s[2][2][PH];
 #pragma unroll
         for (int ef = 0; ef < 2; ++ef)
 #pragma unroll
             for (int mi_ = 0; mi_ < 2; ++mi_)
 #pragma unroll
-                for (int p_ = 0; p_ < 4; ++p_) s[ef][mi_][p_] = f32x2e{0.f, 0.f};
+                for (int p_ = 0; p_ < PH; ++p_) s[ef][mi_][p_] = f32x2e{0.f, 0.f};
 
 #pragma unroll
         for (int mf = 0; mf < 2; ++mf) {   // mf and mf+2 share urow (rows r, r+32)
@@ -270,8 +281,9 @@ pairconv_fwd_kernel(const __bf16* __restrict__ H,
                     bf16x8 uv8 = *reinterpret_cast<const bf16x8*>(
                         u_lds + ((size_t)urow * BLK_E + e) * 8);
 #pragma unroll
-                    for (int p_ = 0; p_ < 4; ++p_) {
-                        f32x2e u2 = {(float)uv8[2 * p_], (float)uv8[2 * p_ + 1]};
+                    for (int p_ = 0; p_ < PH; ++p_) {
+                        const int pp = oh * PH + p_;
+                        f32x2e u2 = {(float)uv8[2 * pp], (float)uv8[2 * pp + 1]};
                         s[ef][0][p_] += rv0 * u2;
                         s[ef][1][p_] += rv1 * u2;
                     }
@@ -285,10 +297,10 @@ pairconv_fwd_kernel(const __bf16* __restrict__ H,
 #pragma unroll
             for (int mi_ = 0; mi_ < 2; ++mi_)
 #pragma unroll
-                for (int p_ = 0; p_ < 4; ++p_) {
+                for (int p_ = 0; p_ < PH; ++p_) {
 #pragma unroll
                     for (int c2 = 0; c2 < 2; ++c2) {
-                        if (2 * p_ + c2 >= O) break;
+                        if (2 * (oh * PH + p_) + c2 >= O) break;
                         float v = s[ef][mi_][p_][c2];
                         v += __shfl_xor(v, 16);
                         v += __shfl_xor(v, 32);
@@ -303,13 +315,15 @@ pairconv_fwd_kernel(const __bf16* __restrict__ H,
                 for (int mi_ = 0; mi_ < 2; ++mi_) {
                     const int moi = sub * BLK_MO + wm * 2 + mi_;
 #pragma unroll
-                    for (int o = 0; o < O; ++o) {
+                    for (int o = 2 * oh * PH;
+                         o < 2 * (oh + 1) * PH && o < O; ++o) {
                         float* p = part + ((size_t)e * (BLK_MO * MB2) + moi) * O + o;
-                        *p += s[ef][mi_][o >> 1][o & 1];
+                        *p += s[ef][mi_][(o >> 1) - oh * PH][o & 1];
                     }
                 }
             }
         }
+        }   // oh
         }   // sub
         __syncthreads();
     }
@@ -359,31 +373,40 @@ static void launch_fwd(const torch::Tensor& H, const torch::Tensor& W,
     const char* wp_env = getenv("SE3_FWD_WP");
     int wp = wp_env ? atoi(wp_env) : ((O == 3 || O == 5) ? 1 : 0);
     if (wp) uu = 0;
+    // ES (epilogue split) — A/B knob for the large orders; see the kernel
+    const char* es_env = getenv("SE3_FWD_ES");
+    int es = es_env ? atoi(es_env) : 0;
+    if (O < 5) es = 0;
+    if (es) uu = 0;
     int ng = mo / (BLK_MO * mb2);
     int coh = (ng % 8 == 0) ? 1 : 0;
     dim3 grid(nmemb * ng);
     size_t lds = 16384 + (size_t)UCHUNK * BLK_E * 8 * 2
                  + (size_t)BLK_E * BLK_MO * mb2 * O * 4;
-#define LAUNCH_FWD(UU, MB2, WPv)                                                             \
-    hipLaunchKernelGGL(HIP_KERNEL_NAME(pairconv_fwd_kernel<O, UU, MB2, WPv>), grid,          \
+#define LAUNCH_FWD(UU, MB2, WPv, ESv)                                                        \
+    hipLaunchKernelGGL(HIP_KERNEL_NAME(pairconv_fwd_kernel<O, UU, MB2, WPv, ESv>), grid,     \
                        dim3(NTHREADS), lds, stream,                                          \
                        reinterpret_cast<const __bf16*>(H.data_ptr()),                        \
                        reinterpret_cast<const __bf16*>(W.data_ptr()),                        \
                        reinterpret_cast<const __bf16*>(Ut.data_ptr()),                       \
                        out.data_ptr<float>(), E, mo, miF, nmemb, coh)  /* W arg = packed P */
     if (mb2 == 2) {
-        if (wp) { LAUNCH_FWD(0, 2, 1); }
+        if (es && wp) { LAUNCH_FWD(0, 2, 1, 1); }
+        else if (es) { LAUNCH_FWD(0, 2, 0, 1); }
+        else if (wp) { LAUNCH_FWD(0, 2, 1, 0); }
         else switch (uu) {
-            case 0: LAUNCH_FWD(0, 2, 0); break;
-            case 1: LAUNCH_FWD(1, 2, 0); break;
-            default: LAUNCH_FWD(2, 2, 0); break;
+            case 0: LAUNCH_FWD(0, 2, 0, 0); break;
+            case 1: LAUNCH_FWD(1, 2, 0, 0); break;
+            default: LAUNCH_FWD(2, 2, 0, 0); break;
         }
     } else {
-        if (wp) { LAUNCH_FWD(0, 1, 1); }
+        if (es && wp) { LAUNCH_FWD(0, 1, 1, 1); }
+        else if (es) { LAUNCH_FWD(0, 1, 0, 1); }
+        else if (wp) { LAUNCH_FWD(0, 1, 1, 0); }
         else switch (uu) {
-            case 0: LAUNCH_FWD(0, 1, 0); break;
-            case 1: LAUNCH_FWD(1, 1, 0); break;
-            default: LAUNCH_FWD(2, 1, 0); break;
+            case 0: LAUNCH_FWD(0, 1, 0, 0); break;
+            case 1: LAUNCH_FWD(1, 1, 0, 0); break;
+            default: LAUNCH_FWD(2, 1, 0, 0); break;
         }
     }
 #undef LAUNCH_FWD
