@@ -174,3 +174,44 @@ def test_equivariance_num_degrees_4_f64_cuda():
         out2 = model(feats, coors, mask, return_type=1) @ R
         diff = (out1 - out2).abs().max().item()
         assert diff < 1e-8, f'degree-4 f64 equivariance on GPU: {diff}'
+
+
+@needs_gpu
+def test_hipgraph_full_step_capture_replay():
+    """Capture one fused train step (fwd+bwd+SGD) in a hipGraph and replay:
+    the mechanism bench.py's single-GPU default relies on. Parameters must
+    keep moving across replays and grads stay finite."""
+    torch.manual_seed(30)
+    model = SE3Transformer(dim=32, heads=2, dim_head=16, depth=1,
+                           num_degrees=2, num_neighbors=6,
+                           attend_self=True).to('cuda')
+    opt = torch.optim.SGD(model.parameters(), lr=1e-3)
+    feats = torch.randn(1, 48, 32, device='cuda')
+    coors = torch.randn(1, 48, 3, device='cuda')
+    mask = torch.ones(1, 48, dtype=torch.bool, device='cuda')
+
+    def step():
+        opt.zero_grad(set_to_none=False)
+        with torch.autocast(device_type='cuda', dtype=torch.bfloat16):
+            out = model(feats, coors, mask, return_type=0)
+            loss = out.float().pow(2).mean()
+        loss.backward()
+        opt.step()
+
+    for _ in range(2):        # allocator warmup
+        step()
+    torch.cuda.synchronize()
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        step()
+    p0 = next(model.parameters())
+    before = p0.detach().clone()
+    g.replay()
+    torch.cuda.synchronize()
+    after1 = p0.detach().clone()
+    g.replay()
+    torch.cuda.synchronize()
+    after2 = p0.detach().clone()
+    assert not torch.equal(before, after1), 'replay 1 did not update params'
+    assert not torch.equal(after1, after2), 'replay 2 did not update params'
+    assert all(torch.isfinite(p).all() for p in model.parameters())
