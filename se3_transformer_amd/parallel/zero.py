@@ -1,0 +1,125 @@
+"""ZeRO-1 optimizer-state sharding over RCCL/xGMI.
+
+The reference has no distributed code at all; the DP layer
+(`DistributedDataParallelSE3`) already gives bucketed gradient all-reduce.
+This adds the next production-scale step: each rank keeps OPTIMIZER STATE
+only for its shard of the parameters (Adam's m/v are 2x the fp32 params —
+at the 18.2B-parameter headline model that is ~146 GB of state, or
+~18 GB/rank sharded 8 ways), steps its shard locally after the gradient
+all-reduce, and broadcasts the updated shard to the other ranks as flat
+per-owner buckets over RCCL.
+
+Composition:
+
+    model = SE3Transformer(...)
+    ddp   = DistributedDataParallelSE3(model)
+    opt   = Zero1Optimizer(model.parameters(), torch.optim.AdamW, lr=1e-4)
+    ...
+    ddp.zero_grad_buffers(); loss.backward(); ddp.finalize(); opt.step()
+
+Works with the gloo backend on CPU for the multi-process tests; with one
+process (or no process group) it degrades to a plain optimizer.
+"""
+from __future__ import annotations
+
+import torch
+import torch.distributed as dist
+
+__all__ = ['Zero1Optimizer']
+
+
+class Zero1Optimizer:
+    """Optimizer-state sharding (ZeRO stage 1).
+
+    Parameters are assigned to ranks greedily by size (largest first onto
+    the currently lightest rank) so the state is balanced; each rank
+    constructs the inner optimizer over its own shard only. `step()` runs
+    the local shard update and then syncs every shard from its owner with
+    one flat broadcast per owner rank (async, overlapped).
+    """
+
+    def __init__(self, params, optim_cls, process_group=None, **optim_kwargs):
+        self.params = [p for p in params if p.requires_grad]
+        self.process_group = process_group
+        if dist.is_available() and dist.is_initialized():
+            self.world = dist.get_world_size(process_group)
+            self.rank = dist.get_rank(process_group)
+        else:
+            self.world, self.rank = 1, 0
+
+        # greedy balanced partition (largest params first)
+        self.owner = {}
+        loads = [0] * self.world
+        for p in sorted(self.params, key=lambda q: -q.numel()):
+            r = min(range(self.world), key=lambda i: loads[i])
+            self.owner[p] = r
+            loads[r] += p.numel()
+        self.shard = [p for p in self.params if self.owner[p] == self.rank]
+        self.opt = optim_cls(self.shard if self.shard else
+                             [torch.nn.Parameter(torch.zeros(1))],
+                             **optim_kwargs)
+        self._shard_empty = not self.shard
+
+        # flat broadcast buffers, one per owner rank (built lazily so they
+        # live on the params' device)
+        self._bcast = None
+
+    def _build_bcast(self):
+        per_owner = [[] for _ in range(self.world)]
+        for p in self.params:
+            per_owner[self.owner[p]].append(p)
+        self._bcast = []
+        for r, plist in enumerate(per_owner):
+            if not plist:
+                self._bcast.append((None, []))
+                continue
+            total = sum(p.numel() for p in plist)
+            buf = torch.empty(total, dtype=plist[0].dtype,
+                              device=plist[0].device)
+            views = []
+            off = 0
+            for p in plist:
+                views.append((p, buf[off: off + p.numel()]))
+                off += p.numel()
+            self._bcast.append((buf, views))
+
+    def zero_grad(self, set_to_none=False):
+        for p in self.params:
+            if p.grad is not None:
+                if set_to_none:
+                    p.grad = None
+                else:
+                    p.grad.zero_()
+
+    @torch.no_grad()
+    def step(self):
+        if not self._shard_empty:
+            self.opt.step()
+        if self.world <= 1:
+            return
+        if self._bcast is None:
+            self._build_bcast()
+        works = []
+        for r, (buf, views) in enumerate(self._bcast):
+            if buf is None:
+                continue
+            if r == self.rank:   # pack the freshly stepped shard
+                for p, v in views:
+                    v.copy_(p.data.reshape(-1))
+            src = (dist.get_global_rank(self.process_group, r)
+                   if self.process_group is not None else r)
+            works.append((r, dist.broadcast(buf, src=src, async_op=True,
+                                            group=self.process_group)))
+        for r, w in works:
+            w.wait()
+        for r, (buf, views) in enumerate(self._bcast):
+            if buf is None or r == self.rank:
+                continue
+            for p, v in views:
+                p.data.copy_(v.view_as(p))
+
+    def state_dict(self):
+        return self.opt.state_dict()
+
+    def load_state_dict(self, sd):
+        self.opt.load_state_dict(sd)
