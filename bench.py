@@ -88,6 +88,12 @@ def main():
                    choices=['none', 'bf16'],
                    help='DP gradient all-reduce dtype (bf16 halves xGMI '
                         'traffic; see parallel/ddp.py for the error bound)')
+    p.add_argument('--optimizer', default='sgd', choices=['sgd', 'adam'])
+    p.add_argument('--zero1', action='store_true',
+                   help='shard optimizer state across ranks (ZeRO-1, '
+                        'parallel/zero.py). At the 18.2B-param headline '
+                        'config Adam state is ~146 GB — it only fits '
+                        'sharded across multiple GPUs.')
     p.add_argument('--preset', default=None, choices=sorted(PRESETS),
                    help='one of the BASELINE.json named configs')
     args = p.parse_args()
@@ -118,7 +124,13 @@ def main():
         grad_compression=args.grad_compression) if world > 1 else None
     runner = ddp if ddp is not None else model
 
-    opt = torch.optim.SGD(model.parameters(), lr=1e-4)
+    optim_cls = torch.optim.AdamW if args.optimizer == 'adam' \
+        else torch.optim.SGD
+    if args.zero1:
+        from se3_transformer_amd.parallel import Zero1Optimizer
+        opt = Zero1Optimizer(model.parameters(), optim_cls, lr=1e-4)
+    else:
+        opt = optim_cls(model.parameters(), lr=1e-4)
 
     # synthetic protein-like point clouds, fixed per rank
     g = torch.Generator(device='cpu').manual_seed(1000 + rank)
@@ -258,7 +270,7 @@ def main():
                 'global_batch': args.batch * world,
                 'parallelism': f'dp{world}',
                 'n_params': n_params,
-                'optimizer': 'sgd',
+                'optimizer': args.optimizer + ('+zero1' if args.zero1 else ''),
                 'preset': args.preset,
             },
         }))
