@@ -1,13 +1,13 @@
-"""Fused pairwise-convolution dispatch (MI355X HIP extension).
+"""Fused-kernel dispatch layer (MI355X HIP extension, se3_transformer_amd._C).
 
-Forward runs the hand-written CDNA4 kernel (csrc/pairconv.hip): a bf16 MFMA
-GEMM over (edges x radial-output-columns) with the basis contraction fused
-into the epilogue — the per-edge radial output R (reference
-se3_transformer_pytorch.py:297-343) never touches HBM.
-
-Backward is computed chunk-wise with library GEMMs (dR re-derived from
-(grad_out, u) on the fly, R re-derived from (H, W) for du) — same
-no-R-materialization principle; to be replaced by dedicated HIP kernels.
+Autograd Functions wrapping the hand-written CDNA4 kernels in csrc/:
+pairwise convolution (fwd + the three dedicated backward kernels + the
+one-pass dual-layout weight pack — the per-edge radial output R of
+reference se3_transformer_pytorch.py:297-343 never touches HBM in either
+direction), the radial trunk, the basis×features precontraction, NormSE3,
+neighbor attention v2 (online softmax, in-kernel rotary, HIP backward)
+and the EGNN higher-type ops. A chunked library-GEMM backward for the
+pairconv remains behind SE3_TORCH_BWD as the parity oracle.
 """
 from __future__ import annotations
 
