@@ -14,6 +14,8 @@ multi-process tests.
 """
 from __future__ import annotations
 
+from contextlib import contextmanager
+
 import torch
 import torch.distributed as dist
 from torch import nn
@@ -168,8 +170,21 @@ class DistributedDataParallelSE3(nn.Module):
             b.work = dist.all_reduce(b.buffer, async_op=True, group=self.process_group)
         b.launched = True
 
+    @contextmanager
+    def no_sync(self):
+        """Gradient accumulation: backwards inside this context accumulate
+        into the flat bucket buffers WITHOUT launching all-reduce (the
+        reference's denoise loop accumulates 16 micro-batches,
+        denoise.py:89). The final backward outside the context reduces the
+        accumulated sum as usual."""
+        self._no_sync = True
+        try:
+            yield
+        finally:
+            self._no_sync = False
+
     def _on_grad(self, p):
-        if not self._comm_active:
+        if not self._comm_active or getattr(self, '_no_sync', False):
             return
         b = self._param_bucket.get(p)
         if b is None or b.launched:

@@ -115,3 +115,48 @@ def test_ddp_bf16_compression_gloo(tmp_path):
     g0 = torch.load(tmp_path / 'bf16_g0.pt')
     g1 = torch.load(tmp_path / 'bf16_g1.pt')
     assert torch.equal(g0, g1)  # both ranks hold the same reduced grads
+
+
+def _accum_worker(rank, world, port, results):
+    import os
+    os.environ['MASTER_ADDR'] = '127.0.0.1'
+    os.environ['MASTER_PORT'] = str(port)
+    torch.distributed.init_process_group('gloo', rank=rank, world_size=world)
+    try:
+        from se3_transformer_amd.parallel import DistributedDataParallelSE3
+        model = _build_model()
+        ddp = DistributedDataParallelSE3(model, bucket_bytes=1 << 16)
+        feats, coors, mask = _make_batch()   # 4 samples total
+        ddp.zero_grad_buffers()
+        # each rank: 2 micro-batches of 1 sample under no_sync + final one
+        my = [rank * 2, rank * 2 + 1]
+        with ddp.no_sync():
+            s = slice(my[0], my[0] + 1)
+            out = ddp(feats[s], coors[s], mask[s], return_type=1)
+            (out.pow(2).mean() / 2).backward()
+        s = slice(my[1], my[1] + 1)
+        out = ddp(feats[s], coors[s], mask[s], return_type=1)
+        (out.pow(2).mean() / 2).backward()
+        ddp.finalize()
+        if rank == 0:
+            results['grads'] = {n: p.grad.clone()
+                                for n, p in model.named_parameters()
+                                if p.grad is not None}
+    finally:
+        torch.distributed.destroy_process_group()
+
+
+def test_ddp_no_sync_grad_accumulation():
+    """2 ranks x (2 accumulated micro-batches of 1 sample) must reproduce
+    the single-process full-batch gradients: no_sync defers the
+    all-reduce, the final backward reduces the accumulated sum."""
+    ref = _single_process_grads()
+    ctx = mp.get_context('spawn')
+    with ctx.Manager() as man:
+        results = man.dict()
+        mp.start_processes(_accum_worker, args=(2, 29873, results), nprocs=2,
+                           join=True, start_method='spawn')
+        grads = dict(results['grads'])
+    for n, g in ref.items():
+        err = (grads[n] - g).abs().max().item()
+        assert err < 1e-5, f'{n}: {err}'
