@@ -55,10 +55,8 @@ class Zero1Optimizer:
             self.owner[p] = r
             loads[r] += p.numel()
         self.shard = [p for p in self.params if self.owner[p] == self.rank]
-        self.opt = optim_cls(self.shard if self.shard else
-                             [torch.nn.Parameter(torch.zeros(1))],
-                             **optim_kwargs)
-        self._shard_empty = not self.shard
+        # a rank can legitimately own nothing (more ranks than params)
+        self.opt = optim_cls(self.shard, **optim_kwargs) if self.shard else None
 
         # flat broadcast buffers, one per owner rank (built lazily so they
         # live on the params' device)
@@ -93,7 +91,7 @@ class Zero1Optimizer:
 
     @torch.no_grad()
     def step(self):
-        if not self._shard_empty:
+        if self.opt is not None:
             self.opt.step()
         if self.world <= 1:
             return
@@ -119,7 +117,8 @@ class Zero1Optimizer:
                 p.data.copy_(v.view_as(p))
 
     def state_dict(self):
-        return self.opt.state_dict()
+        return self.opt.state_dict() if self.opt is not None else {}
 
     def load_state_dict(self, sd):
-        self.opt.load_state_dict(sd)
+        if self.opt is not None:
+            self.opt.load_state_dict(sd)
