@@ -272,6 +272,7 @@ def main():
                 'n_params': n_params,
                 'optimizer': args.optimizer + ('+zero1' if args.zero1 else ''),
                 'preset': args.preset,
+                'hipgraph': bool(graph_mode),
             },
         }))
 
