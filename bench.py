@@ -195,6 +195,7 @@ def main():
     # and is first-try safe.
     graph_mode = (args.graph if args.graph is not None else world == 1) \
         and use_cuda
+    graph_captured = False
     if graph_mode:
         try:
             for _ in range(2):      # allocator warmup before capture
@@ -206,6 +207,7 @@ def main():
             with torch.cuda.graph(g_step):
                 train_step()
             train_step = lambda: g_step.replay()  # noqa: E731
+            graph_captured = True
             if rank == 0:
                 print('[bench] hipGraph capture ok', file=sys.stderr, flush=True)
         except Exception as e:      # pragma: no cover
@@ -272,7 +274,7 @@ def main():
                 'n_params': n_params,
                 'optimizer': args.optimizer + ('+zero1' if args.zero1 else ''),
                 'preset': args.preset,
-                'hipgraph': bool(graph_mode),
+                'hipgraph': graph_captured,
             },
         }))
 
