@@ -127,11 +127,23 @@ class _FusedPairConv(torch.autograd.Function):
         if ext is not None and os.environ.get('SE3_TORCH_BWD') != '1':
             dH = dW = db = dUt = None
             g_t = g16.permute(1, 2, 0).contiguous()    # (mo, O, E) bf16
+            P1 = Pu = None
+            if ctx.packed:
+                P1, Pu = Pdh, Pf
+            elif (need_H or need_u) and hasattr(ext, 'pack_w_both'):
+                # SE3_LOWMEM_PACK path: re-derive both fragment layouts in
+                # one kernel pass from the saved torch-layout W
+                n128 = mo * miF * K
+                Pu = torch.empty(n128, dtype=torch.bfloat16,
+                                 device=H16.device)
+                P1 = torch.empty_like(Pu)
+                ext.pack_w_both(W16, Pu, P1, mo)
             if need_H:
-                P1 = Pdh if ctx.packed else _pack_w_dh(W16, mo, miF)
+                if P1 is None:
+                    P1 = _pack_w_dh(W16, mo, miF)
                 dH = torch.zeros(E, K, dtype=torch.float32, device=H16.device)
                 ext.pairconv_bwd_dh(g_t, Ut16, P1, dH, mo)
-                del P1
+            P1 = None
             if need_W:
                 Ht = H16.t().contiguous()              # (128, E)
                 dW = torch.empty(mo * miF, K, dtype=torch.float32,
@@ -147,7 +159,8 @@ class _FusedPairConv(torch.autograd.Function):
             if need_u:
                 dUt = torch.empty(miF, O, E, dtype=torch.float32,
                                   device=H16.device)
-                Pu = Pf if ctx.packed else _pack_w_fwd(W16, mo, miF)
+                if Pu is None:
+                    Pu = _pack_w_fwd(W16, mo, miF)
                 ext.pairconv_bwd_du(H16, Pu, b16.float().reshape(-1),
                                     g_t, dUt, mo)
             return dH, dW, db, dUt, None

@@ -621,3 +621,32 @@ def test_egnn_kernels_vs_eager():
         err = _rel_err(g1[n].float(), gref[n].float())
         assert err < max(5 * noise, 5e-3), \
             f'egnn kernel grad {n}: {err} (run noise {noise})'
+
+
+@needs_gpu
+def test_lowmem_pack_mode_bitwise_equal():
+    """SE3_LOWMEM_PACK=1 (save torch-layout W, re-pack in backward) must
+    produce bit-identical grads to the default save-both-packs path."""
+    from se3_transformer_amd import SE3Transformer
+
+    def run():
+        torch.manual_seed(17)
+        model = SE3Transformer(dim=64, heads=4, dim_head=16, depth=1,
+                               num_degrees=2, num_neighbors=6).to('cuda')
+        feats = torch.randn(1, 40, 64, device='cuda')
+        coors = torch.randn(1, 40, 3, device='cuda')
+        mask = torch.ones(1, 40, dtype=torch.bool, device='cuda')
+        with torch.autocast(device_type='cuda', dtype=torch.bfloat16):
+            out = model(feats, coors, mask, return_type=0)
+        out.float().pow(2).mean().backward()
+        return {n: p.grad.clone() for n, p in model.named_parameters()
+                if p.grad is not None}
+
+    ref = run()
+    os.environ['SE3_LOWMEM_PACK'] = '1'
+    try:
+        low = run()
+    finally:
+        del os.environ['SE3_LOWMEM_PACK']
+    for n in ref:
+        assert torch.equal(ref[n], low[n]), f'lowmem grad differs: {n}'
