@@ -624,26 +624,31 @@ def test_egnn_kernels_vs_eager():
 
 
 @needs_gpu
-def test_lowmem_pack_mode_bitwise_equal():
+def test_lowmem_pack_mode_matches_default():
     """SE3_LOWMEM_PACK=1 (save torch-layout W, re-pack in backward) must
-    match the default save-both-packs path. The packs themselves are
-    byte-identical (test_pack_w_both_matches_python_permutes); the
-    full-model grads carry the usual atomic-ordering noise (bwd_dh split-K
-    and gather scatter-adds), so a tight tolerance is used."""
-    from se3_transformer_amd import SE3Transformer
+    match the default save-both-packs path at the Function level: the
+    forward kernel and dw/du/db are deterministic (bit-equal expected);
+    dh uses split-K atomics (tight tolerance)."""
+    from se3_transformer_amd.ops.fused import fused_pairconv
+
+    torch.manual_seed(18)
+    device = torch.device('cuda')
+    mo, mi, F_, O, E = 16, 32, 5, 5, 700
+    miF = mi * F_
 
     def run():
-        torch.manual_seed(17)
-        model = SE3Transformer(dim=64, heads=4, dim_head=16, depth=1,
-                               num_degrees=2, num_neighbors=6).to('cuda')
-        feats = torch.randn(1, 40, 64, device='cuda')
-        coors = torch.randn(1, 40, 3, device='cuda')
-        mask = torch.ones(1, 40, dtype=torch.bool, device='cuda')
-        with torch.autocast(device_type='cuda', dtype=torch.bfloat16):
-            out = model(feats, coors, mask, return_type=0)
-        out.float().pow(2).mean().backward()
-        return {n: p.grad.clone() for n, p in model.named_parameters()
-                if p.grad is not None}
+        g = torch.Generator(device).manual_seed(3)
+        H = torch.randn(E, 128, generator=g, device=device,
+                        requires_grad=True)
+        W = torch.randn(mo * miF, 128, generator=g, device=device,
+                        requires_grad=True)
+        bias = torch.randn(mo * miF, generator=g, device=device,
+                           requires_grad=True)
+        Ut = torch.randn(miF, O, E, generator=g, device=device,
+                         requires_grad=True)
+        out = fused_pairconv(H, W, bias, Ut, mo)
+        out.pow(2).mean().backward()
+        return out.detach(), H.grad, W.grad, bias.grad, Ut.grad
 
     ref = run()
     os.environ['SE3_LOWMEM_PACK'] = '1'
@@ -651,8 +656,9 @@ def test_lowmem_pack_mode_bitwise_equal():
         low = run()
     finally:
         del os.environ['SE3_LOWMEM_PACK']
-    for n in ref:
-        err = _rel_err(low[n].float(), ref[n].float())
-        # measured atomic-ordering noise tops out ~1.5e-4 on the trunk
-        # params (dH split-K atomics feeding the radial backward's atomics)
-        assert err < 1e-3, f'lowmem grad differs: {n}: {err}'
+    names = ('out', 'dH', 'dW', 'db', 'dUt')
+    for name, a, b in zip(names, low, ref):
+        if name == 'dH':   # split-K atomics: order noise only
+            assert _rel_err(a, b) < 1e-5, f'{name}'
+        else:
+            assert torch.equal(a, b), f'{name} differs between pack modes'
