@@ -89,6 +89,7 @@ class DistributedDataParallelSE3(nn.Module):
         self.grad_compression = grad_compression
         self.world_size = dist.get_world_size(process_group) if dist.is_initialized() else 1
         self._comm_active = self.world_size > 1 or (force_comm and dist.is_initialized())
+        self._no_sync = False
 
         self._params = [p for p in module.parameters() if p.requires_grad]
         self._buckets = []
@@ -184,7 +185,7 @@ class DistributedDataParallelSE3(nn.Module):
             self._no_sync = False
 
     def _on_grad(self, p):
-        if not self._comm_active or getattr(self, '_no_sync', False):
+        if not self._comm_active or self._no_sync:
             return
         b = self._param_bucket.get(p)
         if b is None or b.launched:
