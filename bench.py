@@ -126,11 +126,18 @@ def main():
 
     optim_cls = torch.optim.AdamW if args.optimizer == 'adam' \
         else torch.optim.SGD
+    opt_kwargs = dict(lr=1e-4)
+    will_graph = (args.graph if args.graph is not None else world == 1) \
+        and use_cuda
+    if args.optimizer == 'adam' and will_graph:
+        # lets AdamW.step() run under hipGraph capture (state tensors stay
+        # on-device instead of python scalars)
+        opt_kwargs['capturable'] = True
     if args.zero1:
         from se3_transformer_amd.parallel import Zero1Optimizer
-        opt = Zero1Optimizer(model.parameters(), optim_cls, lr=1e-4)
+        opt = Zero1Optimizer(model.parameters(), optim_cls, **opt_kwargs)
     else:
-        opt = optim_cls(model.parameters(), lr=1e-4)
+        opt = optim_cls(model.parameters(), **opt_kwargs)
 
     # synthetic protein-like point clouds, fixed per rank
     g = torch.Generator(device='cpu').manual_seed(1000 + rank)
