@@ -53,6 +53,11 @@ def main():
     p.add_argument('--bf16', action='store_true')
     p.add_argument('--zero1', action='store_true',
                    help='shard the Adam state across ranks (ZeRO-1)')
+    p.add_argument('--checkpoint', type=str, default=None,
+                   help='checkpoint path; resumes from it when it exists. '
+                        'Under --zero1 each rank also writes its optimizer '
+                        'shard to <path>.r<rank>')
+    p.add_argument('--save-every', type=int, default=100)
     args = p.parse_args()
 
     rank, world, local_rank = setup_distributed()
@@ -86,6 +91,42 @@ def main():
         optim = Adam(transformer.parameters(), lr=1e-4)
     g = torch.Generator().manual_seed(1000 + rank)   # data shard per rank
 
+    # checkpoint/resume: model weights are replicated (rank 0 writes them);
+    # optimizer state is rank-local under ZeRO-1, so each rank persists its
+    # own shard file next to the main checkpoint
+    start_step = 0
+    shard_path = (f'{args.checkpoint}.r{rank}'
+                  if args.checkpoint and args.zero1 and world > 1 else None)
+    if args.checkpoint and os.path.exists(args.checkpoint):
+        ck = torch.load(args.checkpoint, map_location=device,
+                        weights_only=False)
+        transformer.load_state_dict(ck['model'])
+        if shard_path:
+            if os.path.exists(shard_path):
+                optim.load_state_dict(torch.load(
+                    shard_path, map_location=device, weights_only=False))
+        else:
+            optim.load_state_dict(ck['opt'])
+        start_step = ck['step']
+        if rank == 0:
+            print(f'resumed from {args.checkpoint} at step {start_step}',
+                  flush=True)
+        # advance the data stream so the resumed run sees fresh batches
+        g.manual_seed(1000 + rank + start_step * 100003)
+
+    def save_checkpoint(step):
+        if not args.checkpoint:
+            return
+        if rank == 0:
+            tmp = args.checkpoint + '.tmp'
+            torch.save({'step': step, 'model': transformer.state_dict(),
+                        'opt': {} if shard_path else optim.state_dict()}, tmp)
+            os.replace(tmp, args.checkpoint)   # atomic: never a torn file
+        if shard_path:
+            tmp = shard_path + '.tmp'
+            torch.save(optim.state_dict(), tmp)
+            os.replace(tmp, shard_path)
+
     import contextlib
 
     def micro_batch():
@@ -106,7 +147,7 @@ def main():
         (loss / GRADIENT_ACCUMULATE_EVERY).backward()
         return loss
 
-    for step in range(args.steps):
+    for step in range(start_step, args.steps):
         if ddp is not None:
             ddp.zero_grad_buffers()
         # all-reduce only on the last accumulation micro-batch
@@ -121,6 +162,8 @@ def main():
             print('loss:', loss.item(), flush=True)
         optim.step()
         optim.zero_grad()
+        if (step + 1) % args.save_every == 0 or step + 1 == args.steps:
+            save_checkpoint(step + 1)
 
 
 if __name__ == '__main__':
