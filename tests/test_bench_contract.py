@@ -95,3 +95,35 @@ def test_denoise_checkpoint_resume(tmp_path):
     assert 'resumed from' in r.stdout and 'at step 1' in r.stdout
     # exactly one more optimizer step ran
     assert r.stdout.count('loss:') == 1
+
+
+def test_serve_app_in_process():
+    """examples/serve.py app answers /health and /predict via the in-process
+    test client (no socket)."""
+    import importlib.util
+    spec = importlib.util.spec_from_file_location(
+        'serve_example', os.path.join(REPO, 'examples', 'serve.py'))
+    serve = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(serve)
+    from fastapi.testclient import TestClient
+
+    args = serve.parse_args(['--dim', '16', '--heads', '2', '--dim-head', '8',
+                             '--depth', '1', '--num-neighbors', '4'])
+    client = TestClient(serve.build_app(args))
+
+    r = client.get('/health')
+    assert r.status_code == 200 and r.json()['status'] == 'ok'
+
+    import torch
+    g = torch.Generator().manual_seed(3)
+    feats = torch.randn(12, 16, generator=g).tolist()
+    coors = torch.randn(12, 3, generator=g).tolist()
+    r = client.post('/predict', json={'feats': feats, 'coors': coors})
+    assert r.status_code == 200, r.text
+    body = r.json()
+    assert len(body['output']) == 12 and len(body['output'][0]) == 16
+    assert body['latency_ms'] > 0
+
+    # shape validation is a 422, not a 500
+    r = client.post('/predict', json={'feats': feats, 'coors': coors[:5]})
+    assert r.status_code == 422
