@@ -4,10 +4,11 @@
 The reference library ships no serving path; this is the online-inference
 counterpart to examples/infer.py (which measures offline throughput):
 one resident model per GPU, requests run under `inference_mode` +
-autocast-bf16 through the fused HIP kernel path. A checkpoint saved by
-examples/denoise.py (`{'model': state_dict, ...}`) or a raw state_dict
-can be loaded with --checkpoint; otherwise the model serves random-init
-weights (useful for latency testing only).
+autocast-bf16 through the fused HIP kernel path. --checkpoint loads a
+`{'model': state_dict, ...}` file (the format examples/denoise.py
+writes) or a raw state_dict — the saved architecture must match the
+--dim/--depth/... flags given here; otherwise the model serves
+random-init weights (useful for latency testing only).
 
     python examples/serve.py [--port 8000] [--dim 64 ...] [--checkpoint ck.pt]
 
