@@ -162,7 +162,13 @@ def _qj_t_on(J: int, d_in: int, d_out: int, device, dtype) -> torch.Tensor:
     key = (J, d_in, d_out, _canon_device(device), dtype)
     t = _qj_dev_cache.get(key)
     if t is None:
-        t = _qj_transposed(J, d_in, d_out).to(device=device, dtype=dtype)
+        # fill outside inference mode: a cache populated during a
+        # torch.inference_mode() forward (e.g. a serving request) would
+        # otherwise hold inference tensors that poison any later training
+        # step in the same process ("Inference tensors cannot be saved
+        # for backward")
+        with torch.inference_mode(False):
+            t = _qj_transposed(J, d_in, d_out).to(device=device, dtype=dtype)
         _qj_dev_cache[key] = t
     return t
 
@@ -211,6 +217,13 @@ def _sh_basis_tables(max_degree: int, device):
     hit = _sh_tables_cache.get(key)
     if hit is not None:
         return hit
+    return _build_sh_tables(key, max_degree, device)
+
+
+@torch.inference_mode(False)
+def _build_sh_tables(key, max_degree, device):
+    # built outside inference mode so the cached tables stay usable by
+    # training steps after a torch.inference_mode() forward (see _qj_t_on)
     L = 2 * max_degree
     qparts, meta, layout = [], [], {}
     off_out = off_q = 0

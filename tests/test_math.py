@@ -215,3 +215,24 @@ def test_utils_batched_index_select_masked_mean():
     assert torch.allclose(mm[1], t[1, :1].mean(dim=0), atol=1e-6)
     # functional: input not mutated (reference masked_fill_ mutates — SURVEY §2.4)
     assert torch.isfinite(t).all()
+
+
+def test_basis_cache_survives_inference_mode():
+    """A basis computed under torch.inference_mode (e.g. a serving request)
+    must not poison the module-level Q_J/table caches for later training:
+    the cached device tensors are built with inference mode forced off."""
+    from se3_transformer_amd.ops import basis as basis_mod
+    basis_mod._qj_dev_cache.clear()
+    basis_mod._sh_tables_cache.clear()
+
+    with torch.inference_mode():
+        get_basis(torch.randn(2, 3, 4, 3), 2)           # fills the caches
+        from se3_transformer_amd.ops.basis import get_basis_packed
+        get_basis_packed(torch.randn(2, 3, 4, 3), 2)
+
+    assert all(not t.is_inference() for t in basis_mod._qj_dev_cache.values())
+
+    x = torch.randn(2, 3, 4, 3, requires_grad=True)
+    loss = sum(v.sum() for v in get_basis(x, 2, differentiable=True).values())
+    loss.backward()                                      # raised before the fix
+    assert x.grad is not None and torch.isfinite(x.grad).all()
