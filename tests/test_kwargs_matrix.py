@@ -22,6 +22,9 @@ CASES = {
     'out_fiber_dict': dict(out_fiber_dict={0: 6, 1: 4}, num_degrees=2),
     'edge_tokens': dict(num_edge_tokens=4, edge_dim=3),
     'tokens_positions': dict(num_tokens=11, num_positions=64),
+    'one_headed_kv': dict(one_headed_key_values=True),
+    'causal': dict(causal=True),
+    'input_degrees_2': dict(input_degrees=2, dim_in=(12, 6)),
 }
 
 
@@ -36,6 +39,8 @@ def test_kwarg_combo(name):
     b, n = 1, 16
     if name == 'tokens_positions':
         feats = torch.randint(0, 11, (b, n))
+    elif name == 'input_degrees_2':
+        feats = {'0': torch.randn(b, n, 12, 1), '1': torch.randn(b, n, 6, 3)}
     else:
         feats = torch.randn(b, n, 12)
     coors = torch.randn(b, n, 3)
