@@ -1,0 +1,84 @@
+"""Property-based fuzz tests (hypothesis) for the utility primitives and the
+group-theory identities — randomized complements to the fixed-seed suites."""
+import torch
+from hypothesis import given, settings, strategies as st
+
+from se3_transformer_amd.utils import (batched_index_select, broadcat,
+                                       fast_split, masked_mean)
+
+COMMON = dict(deadline=None, max_examples=25)
+
+
+@settings(**COMMON)
+@given(n=st.integers(1, 40), splits=st.integers(1, 8), d=st.integers(1, 5))
+def test_fast_split_roundtrip(n, splits, d):
+    t = torch.randn(n, d)
+    chunks = list(fast_split(t, splits, dim=0))
+    assert torch.equal(torch.cat(chunks, dim=0), t)
+    # near-equal chunking: sizes differ by at most 1
+    sizes = [c.shape[0] for c in chunks if c.shape[0] > 0]
+    assert max(sizes) - min(sizes) <= 1
+
+
+@settings(**COMMON)
+@given(b=st.integers(1, 3), n=st.integers(2, 12), k=st.integers(1, 6),
+       d=st.integers(1, 4), seed=st.integers(0, 10_000))
+def test_batched_index_select_matches_loop(b, n, k, d, seed):
+    g = torch.Generator().manual_seed(seed)
+    vals = torch.randn(b, n, d, generator=g)
+    idx = torch.randint(0, n, (b, n, k), generator=g)
+    out = batched_index_select(vals, idx, dim=1)
+    for bi in range(b):
+        for i in range(n):
+            for j in range(k):
+                assert torch.equal(out[bi, i, j], vals[bi, idx[bi, i, j]])
+
+
+@settings(**COMMON)
+@given(n=st.integers(1, 10), d=st.integers(1, 5), seed=st.integers(0, 10_000))
+def test_masked_mean_matches_manual(n, d, seed):
+    g = torch.Generator().manual_seed(seed)
+    t = torch.randn(2, n, d, generator=g)
+    mask = torch.rand(2, n, generator=g) > 0.4
+    out = masked_mean(t.clone(), mask, dim=1)
+    for bi in range(2):
+        sel = t[bi][mask[bi]]
+        want = sel.mean(dim=0) if len(sel) else torch.zeros(d)
+        assert torch.allclose(out[bi], want, atol=1e-6)
+
+
+@settings(**COMMON)
+@given(a=st.integers(1, 4), b=st.integers(1, 4), seed=st.integers(0, 10_000))
+def test_broadcat_matches_expand_cat(a, b, seed):
+    g = torch.Generator().manual_seed(seed)
+    x = torch.randn(a, 1, 3, generator=g)
+    y = torch.randn(1, b, 2, generator=g)
+    out = broadcat([x, y], dim=-1)
+    want = torch.cat([x.expand(a, b, 3), y.expand(a, b, 2)], dim=-1)
+    assert torch.equal(out, want)
+
+
+@settings(deadline=None, max_examples=15)
+@given(l=st.integers(0, 4),
+       angles=st.tuples(*[st.floats(-3.1, 3.1, allow_nan=False)] * 3),
+       seed=st.integers(0, 10_000))
+def test_wigner_sh_identity_random_rotations(l, angles, seed):
+    """Y_l(R x) = D_l(R) Y_l(x) for random rotations and points — the
+    defining identity, evaluated with `_sh_at_points` (the standard-frame
+    SH wigner_d is solved against; `sh_packed_from_cartesian` is a
+    DIFFERENT frame — it applies the reference's std→SH axis permutation,
+    whose equivariance contract is covered by the model-level tests).
+    Fixed-angle versions live in test_math.py."""
+    from se3_transformer_amd.ops.wigner import _sh_at_points, rot, wigner_d
+
+    a, b, c = angles
+    R = rot(torch.tensor(a, dtype=torch.float64),
+            torch.tensor(b, dtype=torch.float64),
+            torch.tensor(c, dtype=torch.float64))
+    D = wigner_d(l, a, b, c)
+    g = torch.Generator().manual_seed(seed)
+    x = torch.randn(8, 3, dtype=torch.float64, generator=g)
+    x = x / x.norm(dim=-1, keepdim=True)
+    y = _sh_at_points(l, x)
+    y_rot = _sh_at_points(l, x @ R.t())
+    assert torch.allclose(y_rot, y @ D.t(), atol=1e-10)
