@@ -82,3 +82,41 @@ def test_wigner_sh_identity_random_rotations(l, angles, seed):
     y = _sh_at_points(l, x)
     y_rot = _sh_at_points(l, x @ R.t())
     assert torch.allclose(y_rot, y @ D.t(), atol=1e-10)
+
+
+@settings(deadline=None, max_examples=10)
+@given(pair=st.sampled_from([(0, 1), (1, 1), (1, 2), (2, 2), (2, 1)]),
+       angles=st.tuples(*[st.floats(-3.1, 3.1, allow_nan=False)] * 3),
+       seed=st.integers(0, 10_000))
+def test_basis_equivariance_random_rotations(pair, angles, seed):
+    """K(r @ R) = D_out K(r) D_in^T for random rotations — the hypothesis
+    version of test_math.py::test_basis_kernel_equivariance (same frame
+    bookkeeping: row-convention rotation + the std→SH axis permutation)."""
+    from se3_transformer_amd.ops.basis import get_basis_packed
+    from se3_transformer_amd.ops.wigner import rot, wigner_d_from_matrix
+    from se3_transformer_amd.utils import torch_default_dtype
+
+    d_in, d_out = pair
+    with torch_default_dtype(torch.float64):
+        a, b, c = angles
+        R3 = rot(torch.tensor(a, dtype=torch.float64),
+                 torch.tensor(b, dtype=torch.float64),
+                 torch.tensor(c, dtype=torch.float64))
+        g = torch.Generator().manual_seed(seed)
+        r = torch.randn(8, 3, dtype=torch.float64, generator=g)
+        f = 2 * min(d_in, d_out) + 1
+        w = torch.randn(f, dtype=torch.float64, generator=g)
+
+        def kernel(rr):
+            bas = get_basis_packed(rr, 2)[(d_in, d_out)].double()
+            return (bas * w).sum(-1)
+
+        k1 = kernel(r @ R3)
+        k0 = kernel(r)
+        P = torch.tensor([[0., 0., 1.], [1., 0., 0.], [0., 1., 0.]],
+                         dtype=torch.float64)
+        m_perm = P @ R3.t() @ P.t()
+        d_o = wigner_d_from_matrix(d_out, m_perm)
+        d_i = wigner_d_from_matrix(d_in, m_perm)
+        rhs = torch.einsum('oi,nij,pj->nop', d_o, k0, d_i)
+        assert (k1 - rhs).abs().max() < 1e-8
