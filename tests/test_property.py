@@ -6,7 +6,9 @@ from hypothesis import given, settings, strategies as st
 from se3_transformer_amd.utils import (batched_index_select, broadcat,
                                        fast_split, masked_mean)
 
-COMMON = dict(deadline=None, max_examples=25)
+# derandomize: deterministic example set per test (the CI gate must not
+# flake; broad randomized sweeps were run during development)
+COMMON = dict(deadline=None, max_examples=25, derandomize=True)
 
 
 @settings(**COMMON)
@@ -58,7 +60,7 @@ def test_broadcat_matches_expand_cat(a, b, seed):
     assert torch.equal(out, want)
 
 
-@settings(deadline=None, max_examples=15)
+@settings(deadline=None, max_examples=15, derandomize=True)
 @given(l=st.integers(0, 4),
        angles=st.tuples(*[st.floats(-3.1, 3.1, allow_nan=False)] * 3),
        seed=st.integers(0, 10_000))
@@ -84,7 +86,7 @@ def test_wigner_sh_identity_random_rotations(l, angles, seed):
     assert torch.allclose(y_rot, y @ D.t(), atol=1e-10)
 
 
-@settings(deadline=None, max_examples=10)
+@settings(deadline=None, max_examples=10, derandomize=True)
 @given(pair=st.sampled_from([(0, 1), (1, 1), (1, 2), (2, 2), (2, 1)]),
        angles=st.tuples(*[st.floats(-3.1, 3.1, allow_nan=False)] * 3),
        seed=st.integers(0, 10_000))
@@ -124,7 +126,7 @@ def test_basis_equivariance_random_rotations(pair, angles, seed):
         assert (k1 - rhs).abs().max() < 1e-6
 
 
-@settings(deadline=None, max_examples=15)
+@settings(deadline=None, max_examples=15, derandomize=True)
 @given(l=st.integers(0, 4),
        A=st.tuples(*[st.floats(-3.1, 3.1, allow_nan=False)] * 6))
 def test_compose_representation_homomorphism(l, A):
