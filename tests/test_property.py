@@ -140,3 +140,32 @@ def test_compose_representation_homomorphism(l, A):
         lhs = irr_repr(l, a, b, c)
         rhs = irr_repr(l, *A[:3]) @ irr_repr(l, *A[3:])
         assert (lhs - rhs).abs().max() < 1e-6
+
+
+def test_model_invariants_neighbor_mask_noop_and_permutation():
+    """Two end-to-end invariants of the full model (f64, eager):
+    1. an all-True neighbor_mask must not change the output;
+    2. permuting the input points permutes the output identically
+       (kNN graph construction, gathers and attention are all
+       permutation-equivariant; ties have measure zero at random f64
+       coords). Both hold EXACTLY (identical op sequence)."""
+    torch.manual_seed(0)
+    from se3_transformer_amd import SE3Transformer
+    model = SE3Transformer(dim=16, heads=2, dim_head=8, depth=2,
+                           num_degrees=2, num_neighbors=4,
+                           attend_self=True).double()
+    n = 12
+    feats = torch.randn(1, n, 16, dtype=torch.float64)
+    coors = torch.randn(1, n, 3, dtype=torch.float64)
+    mask = torch.ones(1, n, dtype=torch.bool)
+
+    out = model(feats, coors, mask, return_type=0)
+
+    out_nm = model(feats, coors, mask, return_type=0,
+                   neighbor_mask=torch.ones(1, n, n, dtype=torch.bool))
+    assert torch.equal(out, out_nm)
+
+    perm = torch.randperm(n)
+    out_p = model(feats[:, perm], coors[:, perm], mask[:, perm],
+                  return_type=0)
+    assert torch.allclose(out_p, out[:, perm], atol=1e-12)
