@@ -169,3 +169,25 @@ def test_model_invariants_neighbor_mask_noop_and_permutation():
     out_p = model(feats[:, perm], coors[:, perm], mask[:, perm],
                   return_type=0)
     assert torch.allclose(out_p, out[:, perm], atol=1e-12)
+
+
+def test_model_translation_invariance():
+    """The T in SE(3): translating all coordinates must leave both the
+    type-0 output and the type-1 output unchanged to f64 rounding (everything
+    downstream consumes relative positions only)."""
+    torch.manual_seed(1)
+    from se3_transformer_amd import SE3Transformer
+    model = SE3Transformer(dim=16, heads=2, dim_head=8, depth=2,
+                           num_degrees=2, output_degrees=2,
+                           num_neighbors=4, attend_self=True).double()
+    n = 12
+    feats = torch.randn(1, n, 16, dtype=torch.float64)
+    coors = torch.randn(1, n, 3, dtype=torch.float64)
+    mask = torch.ones(1, n, dtype=torch.bool)
+    t = torch.tensor([12.3, -4.5, 0.71], dtype=torch.float64)
+
+    out0 = model(feats, coors, mask)
+    out1 = model(feats, coors + t, mask)
+    # exact mathematically; (a+t)-(b+t) vs a-b differs by f64 rounding only
+    assert torch.allclose(out0['0'], out1['0'], atol=1e-10)
+    assert torch.allclose(out0['1'], out1['1'], atol=1e-10)
