@@ -191,3 +191,27 @@ def test_model_translation_invariance():
     # exact mathematically; (a+t)-(b+t) vs a-b differs by f64 rounding only
     assert torch.allclose(out0['0'], out1['0'], atol=1e-10)
     assert torch.allclose(out0['1'], out1['1'], atol=1e-10)
+
+
+def test_causal_prefix_independence():
+    """With causal=True, the output at position i must not depend on any
+    later point: perturbing the suffix (both feats and coors) leaves the
+    prefix outputs unchanged (reference causal semantics, :1264-1268)."""
+    torch.manual_seed(2)
+    from se3_transformer_amd import SE3Transformer
+    model = SE3Transformer(dim=16, heads=2, dim_head=8, depth=2,
+                           num_degrees=2, num_neighbors=4,
+                           causal=True, attend_self=True).double()
+    n, cut = 12, 6
+    feats = torch.randn(1, n, 16, dtype=torch.float64)
+    coors = torch.randn(1, n, 3, dtype=torch.float64)
+    mask = torch.ones(1, n, dtype=torch.bool)
+
+    out0 = model(feats, coors, mask, return_type=0)
+    feats2, coors2 = feats.clone(), coors.clone()
+    feats2[:, cut:] = torch.randn_like(feats2[:, cut:])
+    coors2[:, cut:] = torch.randn_like(coors2[:, cut:]) * 3
+    out1 = model(feats2, coors2, mask, return_type=0)
+    assert torch.allclose(out0[:, :cut], out1[:, :cut], atol=1e-12)
+    # sanity: the suffix DID change
+    assert not torch.allclose(out0[:, cut:], out1[:, cut:], atol=1e-3)
