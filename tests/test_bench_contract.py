@@ -127,3 +127,23 @@ def test_serve_app_in_process():
     # shape validation is a 422, not a 500
     r = client.post('/predict', json={'feats': feats, 'coors': coors[:5]})
     assert r.status_code == 422
+
+
+def test_bench_preset_with_explicit_override():
+    """Preset values fill defaults but explicit flags win (bench.py preset
+    precedence): --preset qm9 sets points=29/batch=16, an explicit --batch 2
+    overrides the preset's 16."""
+    import json
+    import subprocess
+    import sys
+    r = subprocess.run(
+        [sys.executable, 'bench.py', '--preset', 'qm9', '--steps', '1',
+         '--warmup', '0', '--batch', '2', '--depth', '1'],
+        capture_output=True, text=True, cwd=REPO, timeout=600)
+    assert r.returncode == 0, r.stderr[-1500:]
+    d = json.loads([l for l in r.stdout.splitlines() if l.startswith('{')][-1])
+    cfg = d['config']
+    assert cfg['points'] == 29          # from the preset
+    assert cfg['global_batch'] == 2     # explicit flag beat the preset's 16
+    assert cfg['depth'] == 1            # explicit flag beat the preset's 4
+    assert d['metric'].endswith('preset=qm9')
