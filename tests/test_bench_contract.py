@@ -147,3 +147,47 @@ def test_bench_preset_with_explicit_override():
     assert cfg['global_batch'] == 2     # explicit flag beat the preset's 16
     assert cfg['depth'] == 1            # explicit flag beat the preset's 4
     assert d['metric'].endswith('preset=qm9')
+
+
+def test_serve_loads_checkpoint(tmp_path):
+    """serve.py --checkpoint actually loads the weights: predictions from a
+    server started on a saved checkpoint differ from random init and equal
+    a direct forward of the checkpointed model."""
+    import importlib.util
+    import torch
+    spec = importlib.util.spec_from_file_location(
+        'serve_example2', os.path.join(REPO, 'examples', 'serve.py'))
+    serve = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(serve)
+    from fastapi.testclient import TestClient
+
+    flags = ['--dim', '16', '--heads', '2', '--dim-head', '8',
+             '--depth', '1', '--num-neighbors', '4']
+    args = serve.parse_args(flags)
+
+    # make a model with DIFFERENT weights than serve's seed-0 init
+    # (build_model seeds itself, so perturb after construction)
+    ref = serve.build_model(serve.parse_args(flags), torch.device('cpu'))
+    with torch.no_grad():
+        for p_ in ref.parameters():
+            p_.add_(torch.randn_like(p_) * 0.05)
+    ck = tmp_path / 'ck.pt'
+    torch.save({'model': ref.state_dict(), 'step': 7}, ck)
+
+    g = torch.Generator().manual_seed(9)
+    feats = torch.randn(12, 16, generator=g)
+    coors = torch.randn(12, 3, generator=g)
+    body = {'feats': feats.tolist(), 'coors': coors.tolist()}
+
+    plain = TestClient(serve.build_app(args)).post('/predict', json=body)
+    loaded = TestClient(serve.build_app(
+        serve.parse_args(flags + ['--checkpoint', str(ck)]))) \
+        .post('/predict', json=body)
+    assert plain.status_code == 200 and loaded.status_code == 200
+    out_loaded = torch.tensor(loaded.json()['output'])
+    assert not torch.allclose(torch.tensor(plain.json()['output']),
+                              out_loaded)                   # weights changed
+    with torch.inference_mode():
+        want = ref(feats.unsqueeze(0), coors.unsqueeze(0),
+                   torch.ones(1, 12, dtype=torch.bool), return_type=0)
+    assert torch.allclose(out_loaded, want.squeeze(0), atol=1e-5)
